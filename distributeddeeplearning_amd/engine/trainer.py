@@ -1,0 +1,223 @@
+"""Training engine: the reference's per-rank hot loop as a
+class.  One process per GPU; backward overlaps bucketed all-reduce; the only
+sync point is ``DistributedOptimizer.synchronize()`` inside ``step``."""
+from __future__ import annotations
+
+import math
+import time
+from typing import Iterator, Optional, Tuple
+
+import torch
+
+from .. import models as model_zoo
+from ..config import TrainConfig
+from ..data.synthetic import synthetic_loader
+from ..ops import FusedSGD, cast_model_
+from ..ops import functional as DF
+from ..parallel import DistributedOptimizer, api
+from . import checkpoint as ckpt
+from .loss_scaler import DynamicLossScaler
+from .metrics import MetricLogger, StepTimer, log, setup_logging
+from .schedule import WarmupPiecewiseLR
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, steps_per_epoch: Optional[int] = None):
+        self.cfg = cfg
+        api.init(cfg.comm)
+        self.rank, self.world = api.rank(), api.size()
+        setup_logging(self.rank)
+        torch.manual_seed(cfg.seed + self.rank)
+
+        self.device = cfg.resolved_device()
+        if self.device.type == "cuda":
+            torch.cuda.set_device(api.local_rank())
+            self.device = torch.device("cuda", api.local_rank())
+
+        dtype = cfg.torch_dtype()
+        model = model_zoo.build_model(
+            cfg.model, num_classes=cfg.num_classes,
+            zero_init_residual=cfg.zero_init_residual,
+        ) if cfg.model.startswith("resnet") else model_zoo.build_model(
+            cfg.model, num_classes=cfg.num_classes)
+        model = model.to(self.device)
+        cast_model_(model, dtype, channels_last=cfg.channels_last)
+        self.model = model
+        self.dtype = dtype
+
+        # steps/epoch: synthetic pretends a 1.28 M-image epoch like ImageNet
+        global_batch = cfg.batch_size * self.world
+        self.steps_per_epoch = steps_per_epoch or max(1, 1_281_167 // global_batch)
+        self.global_batch = global_batch
+
+        fp16 = dtype == torch.float16
+        self.scaler = DynamicLossScaler(
+            init_scale=cfg.loss_scale if cfg.loss_scale > 0 else 2.0 ** 16,
+            enabled=fp16,
+        )
+        if cfg.loss_scale > 0:
+            self.scaler.growth_interval = 10 ** 9  # static scale
+
+        inner = FusedSGD(
+            model.parameters(), lr=cfg.base_lr, momentum=cfg.momentum,
+            weight_decay=cfg.weight_decay, lars=cfg.lars,
+            lars_eta=cfg.lars_eta, grad_scale=self.scaler.scale,
+        )
+        self.opt = DistributedOptimizer(
+            inner, named_parameters=model.named_parameters(),
+            backward_passes_per_step=cfg.backward_passes_per_step,
+            bucket_bytes=int(cfg.bucket_mb * (1 << 20)),
+        )
+        self.lr_sched = WarmupPiecewiseLR(
+            self.opt, cfg.base_lr, global_batch, self.steps_per_epoch,
+            cfg.warmup_epochs, cfg.lr_decay_epochs, cfg.lr_decay_factor,
+        )
+        self.metrics = MetricLogger(cfg.metrics_jsonl, self.rank)
+        self.global_step = 0
+        self.epoch = 0
+
+        if cfg.resume:
+            payload = ckpt.load(cfg.resume, self.model, self.opt, self.scaler,
+                                device=self.device)
+            self.epoch = payload.get("epoch", 0)
+            self.global_step = payload.get("step", 0)
+            log.info("resumed from %s at epoch %d step %d", cfg.resume,
+                     self.epoch, self.global_step)
+        elif self.world > 1:
+            # C2/C3: one fused broadcast each of params and optimizer state
+            api.broadcast_parameters(self.model.state_dict(), root_rank=0)
+            self.opt.optimizer.refresh_master()
+
+        self._loader: Optional[Iterator] = None
+
+    # ------------------------------------------------------------------
+
+    def data_iter(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        if self._loader is None:
+            if self.cfg.data == "synthetic":
+                self._loader = synthetic_loader(
+                    self.cfg.batch_size, self.cfg.image_size,
+                    self.cfg.num_classes, self.dtype, self.device,
+                    seed=self.cfg.seed + self.rank,
+                    channels_last=self.cfg.channels_last,
+                )
+            else:
+                from ..data.imagenet import imagenet_loader
+                self._loader = imagenet_loader(
+                    self.cfg.data, self.cfg.batch_size, self.cfg.image_size,
+                    self.dtype, self.device, rank=self.rank, world=self.world,
+                    channels_last=self.cfg.channels_last,
+                )
+        return self._loader
+
+    def train_step(self, x, y) -> torch.Tensor:
+        lr = self.lr_sched.step(self.global_step)
+        logits = self.model(x)
+        loss = DF.softmax_cross_entropy(logits, y, self.cfg.label_smoothing)
+        self.scaler.scale_loss(loss).backward()
+        if self.scaler.enabled:
+            self.opt.synchronize()
+            flats = [b.flat for b in self.opt.buckets.buckets]
+            self.opt.optimizer.grad_scale = self.scaler.scale
+            if self.scaler.check_and_update(flats):
+                self.opt.step()
+            # on overflow: skip the update entirely (standard AMP)
+        else:
+            self.opt.step()
+        self.opt.zero_grad()
+        self.global_step += 1
+        return loss
+
+    def train_epoch(self, steps: Optional[int] = None) -> float:
+        self.model.train()
+        steps = steps or self.steps_per_epoch
+        it = self.data_iter()
+        timer = StepTimer(self.device)
+        last_loss = float("nan")
+        for i in range(steps):
+            x, y = next(it)
+            loss = self.train_step(x, y)
+            timer.add(self.global_batch)
+            if (i + 1) % self.cfg.log_interval == 0:
+                last_loss = float(loss.detach().float())
+                rate = timer.rate()
+                log.info("epoch %d step %d/%d loss %.4f lr %.4f %.1f images/sec",
+                         self.epoch, i + 1, steps, last_loss,
+                         self.lr_sched.last_lr, rate)
+                self.metrics.write(step=self.global_step, epoch=self.epoch,
+                                   loss=last_loss, lr=self.lr_sched.last_lr,
+                                   images_per_sec=rate,
+                                   loss_scale=self.scaler.scale)
+                timer.reset()
+        self.epoch += 1
+        return last_loss
+
+    def fit(self) -> None:
+        cfg = self.cfg
+        while self.epoch < cfg.epochs:
+            self.train_epoch(cfg.steps)
+            if cfg.checkpoint_dir and (self.epoch % cfg.checkpoint_interval_epochs == 0):
+                path = f"{cfg.checkpoint_dir}/{cfg.model}_epoch{self.epoch}.pt"
+                ckpt.save(path, self.model, self.opt, self.epoch,
+                          self.global_step, self.scaler)
+                log.info("checkpoint: %s", path)
+            if cfg.steps:  # bench/CI mode: bounded total work
+                break
+
+    # ------------------------------------------------------------------
+
+    @torch.no_grad()
+    def evaluate(self, loader, max_batches: Optional[int] = None):
+        from .metrics import average_metric
+        self.model.eval()
+        correct = total = 0
+        loss_sum = 0.0
+        for bi, (x, y) in enumerate(loader):
+            if max_batches and bi >= max_batches:
+                break
+            x = x.to(self.device, self.dtype)
+            y = y.to(self.device)
+            if self.cfg.channels_last and x.dim() == 4:
+                x = x.contiguous(memory_format=torch.channels_last)
+            logits = self.model(x)
+            loss_sum += float(DF.softmax_cross_entropy(logits, y).float())
+            correct += int((logits.float().argmax(1) == y).sum())
+            total += y.numel()
+        top1 = correct / max(1, total)
+        # C4: metric averaging across ranks
+        top1 = average_metric(top1, device="cpu")
+        self.model.train()
+        return {"top1": top1, "val_loss": loss_sum / max(1, total)}
+
+    # ------------------------------------------------------------------
+
+    def bench(self, steps: int, warmup: int) -> dict:
+        """Timed benchmark: W untimed warmups, barrier+sync bracket, K timed
+        steps, MAX elapsed over ranks → whole-job images/sec."""
+        self.model.train()
+        it = self.data_iter()
+        for _ in range(warmup):
+            x, y = next(it)
+            self.train_step(x, y)
+        if self.world > 1:
+            api.barrier()
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            x, y = next(it)
+            self.train_step(x, y)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        elapsed = time.perf_counter() - t0
+        if self.world > 1:
+            t = torch.tensor([elapsed], dtype=torch.float64)
+            # MAX over ranks: a job is as fast as its slowest rank
+            import torch.distributed as torch_dist
+            if torch_dist.is_initialized():
+                torch_dist.all_reduce(t, op=torch_dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+        ms_per_step = elapsed / steps * 1000.0
+        images_per_sec = self.global_batch * steps / elapsed
+        return {"images_per_sec": images_per_sec, "ms_per_step": ms_per_step,
+                "elapsed_s": elapsed}

@@ -1,0 +1,292 @@
+"""Functional op surface with two implementations per op:
+
+* GPU (``tensor.is_cuda``): hand-written HIP/CDNA4 kernels via ``_C_kernels``
+  wrapped in autograd Functions.  NHWC physical layout (torch
+  ``channels_last``), bf16/fp16/fp32 compute.  Missing extension on a GPU
+  machine raises (no silent eager fallback).
+* CPU: plain fp32 PyTorch composition.  This is both the tier-0 test path
+  (BASELINE config 1) and the numerical reference the HIP kernels are
+  validated against (SURVEY.md §4.3 Tier 1).
+
+Reference call sites these ops replace: the cuDNN/cuBLAS launches issued by
+torchvision resnet50 inside ``HorovodPytorch/src/imagenet_pytorch_horovod.py``
+(SURVEY.md §2.3 table K1–K15).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from ._ext import load_kernels_ext
+
+
+def _pair(v):
+    return (v, v) if isinstance(v, int) else tuple(v)
+
+
+def to_channels_last(t: torch.Tensor) -> torch.Tensor:
+    if t.dim() == 4:
+        return t.contiguous(memory_format=torch.channels_last)
+    return t.contiguous()
+
+
+# --------------------------------------------------------------------------
+# Conv2d (K1/K2/K3: implicit-GEMM NHWC, MFMA)
+# --------------------------------------------------------------------------
+
+class _Conv2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding, relu):
+        ext = load_kernels_ext()
+        x = to_channels_last(x)
+        weight = to_channels_last(weight)
+        y = ext.conv_fwd(x, weight, bias if bias is not None else None,
+                         stride[0], stride[1], padding[0], padding[1], relu)
+        if relu:
+            ctx.save_for_backward(x, weight, y)
+        else:
+            ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.padding = padding
+        ctx.has_bias = bias is not None
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_kernels_ext()
+        if ctx.relu:
+            x, weight, y = ctx.saved_tensors
+            dy = ext.relu_bwd(to_channels_last(dy), y)
+        else:
+            x, weight = ctx.saved_tensors
+            dy = to_channels_last(dy)
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv_dgrad(dy, weight, list(x.shape),
+                                ctx.stride[0], ctx.stride[1],
+                                ctx.padding[0], ctx.padding[1])
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv_wgrad(dy, x, list(weight.shape),
+                                ctx.stride[0], ctx.stride[1],
+                                ctx.padding[0], ctx.padding[1])
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.sum(dim=(0, 2, 3))
+        return dx, dw, db, None, None, None
+
+
+def conv2d(x, weight, bias=None, stride=1, padding=0, relu=False):
+    stride, padding = _pair(stride), _pair(padding)
+    if x.is_cuda:
+        return _Conv2dFn.apply(x, weight, bias, stride, padding, relu)
+    y = F.conv2d(x, weight, bias, stride, padding)
+    return F.relu(y) if relu else y
+
+
+# --------------------------------------------------------------------------
+# Fused BatchNorm (+ residual add) (+ ReLU)   (K4/K5/K14)
+# --------------------------------------------------------------------------
+
+class _BnAddReluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var,
+                momentum, eps, relu, residual):
+        ext = load_kernels_ext()
+        x = to_channels_last(x)
+        if residual is not None:
+            residual = to_channels_last(residual)
+        y, mean, rstd = ext.bn_relu_fwd(
+            x, weight, bias, running_mean, running_var,
+            float(momentum), float(eps), relu,
+            residual if residual is not None else None,
+        )
+        ctx.save_for_backward(x, weight, mean, rstd, y)
+        ctx.relu = relu
+        ctx.has_residual = residual is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_kernels_ext()
+        x, weight, mean, rstd, y = ctx.saved_tensors
+        dy = to_channels_last(dy)
+        dx, dw, db, dres = ext.bn_relu_bwd(
+            dy, x, weight, mean, rstd, y, ctx.relu, ctx.has_residual)
+        return (dx, dw, db, None, None, None, None, None,
+                dres if ctx.has_residual else None)
+
+
+def bn_add_relu(x, weight, bias, running_mean, running_var, training,
+                momentum=0.1, eps=1e-5, relu=True, residual=None):
+    """y = relu?(bn(x) + residual?) — single fused kernel each direction.
+
+    The reference issues separate cuDNN BN and ReLU (and a separate add for
+    the bottleneck exit); on MI355X the op is HBM-bound so fusing saves two
+    full activation round-trips."""
+    if x.is_cuda:
+        if training:
+            return _BnAddReluFn.apply(x, weight, bias, running_mean,
+                                      running_var, momentum, eps, relu, residual)
+        ext = load_kernels_ext()
+        return ext.bn_relu_infer(to_channels_last(x), weight, bias,
+                                 running_mean, running_var, float(eps), relu,
+                                 to_channels_last(residual) if residual is not None else None)
+    # CPU reference
+    y = F.batch_norm(x, running_mean, running_var, weight, bias,
+                     training, momentum, eps)
+    if residual is not None:
+        y = y + residual
+    if relu:
+        y = F.relu(y)
+    return y
+
+
+# --------------------------------------------------------------------------
+# Pooling (K6/K7)
+# --------------------------------------------------------------------------
+
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        ext = load_kernels_ext()
+        x = to_channels_last(x)
+        y, idx = ext.maxpool_fwd(x, kernel[0], kernel[1], stride[0], stride[1],
+                                 padding[0], padding[1])
+        ctx.save_for_backward(idx)
+        ctx.x_shape = list(x.shape)
+        ctx.params = (kernel, stride, padding)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_kernels_ext()
+        (idx,) = ctx.saved_tensors
+        kernel, stride, padding = ctx.params
+        dx = ext.maxpool_bwd(to_channels_last(dy), idx, ctx.x_shape,
+                             kernel[0], kernel[1], stride[0], stride[1],
+                             padding[0], padding[1])
+        return dx, None, None, None
+
+
+def max_pool2d(x, kernel_size, stride=None, padding=0):
+    kernel = _pair(kernel_size)
+    stride = _pair(stride) if stride is not None else kernel
+    padding = _pair(padding)
+    if x.is_cuda:
+        return _MaxPoolFn.apply(x, kernel, stride, padding)
+    return F.max_pool2d(x, kernel, stride, padding)
+
+
+class _GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = load_kernels_ext()
+        x = to_channels_last(x)
+        ctx.x_shape = list(x.shape)
+        return ext.gap_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_kernels_ext()
+        return ext.gap_bwd(dy.contiguous(), ctx.x_shape)
+
+
+def global_avg_pool(x):
+    """[N,C,H,W] -> [N,C] mean over H,W."""
+    if x.is_cuda:
+        return _GlobalAvgPoolFn.apply(x)
+    return x.mean(dim=(2, 3))
+
+
+# --------------------------------------------------------------------------
+# Linear (K8: plain library GEMM — rocBLAS/hipBLASLt via torch.matmul)
+# --------------------------------------------------------------------------
+
+def linear(x, weight, bias=None):
+    return F.linear(x, weight, bias)
+
+
+# --------------------------------------------------------------------------
+# Fused softmax + cross-entropy (K9)
+# --------------------------------------------------------------------------
+
+class _SoftmaxXentFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target, label_smoothing):
+        ext = load_kernels_ext()
+        losses, probs = ext.softmax_xent_fwd(logits.contiguous(), target,
+                                             float(label_smoothing))
+        ctx.save_for_backward(probs, target)
+        ctx.label_smoothing = label_smoothing
+        ctx.logits_dtype = logits.dtype
+        return losses.mean()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = load_kernels_ext()
+        probs, target = ctx.saved_tensors
+        dlogits = ext.softmax_xent_bwd(probs, target,
+                                       float(ctx.label_smoothing),
+                                       dloss / probs.shape[0])
+        return dlogits.to(ctx.logits_dtype), None, None
+
+
+def softmax_cross_entropy(logits, target, label_smoothing: float = 0.0):
+    if logits.is_cuda:
+        return _SoftmaxXentFn.apply(logits, target, label_smoothing)
+    return F.cross_entropy(logits.float(), target, label_smoothing=label_smoothing)
+
+
+# --------------------------------------------------------------------------
+# Dropout (K15, VGG path)
+# --------------------------------------------------------------------------
+
+class _DropoutFn(torch.autograd.Function):
+    _seed_counter = [0]
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = load_kernels_ext()
+        y, mask = ext.dropout_fwd(x.contiguous(), float(p), int(seed))
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = load_kernels_ext()
+        (mask,) = ctx.saved_tensors
+        return ext.dropout_bwd(dy.contiguous(), mask, float(ctx.p)), None, None
+
+
+def dropout(x, p=0.5, training=True, seed: Optional[int] = None):
+    if not training or p == 0.0:
+        return x
+    if x.is_cuda:
+        if seed is None:
+            _DropoutFn._seed_counter[0] += 1
+            seed = torch.initial_seed() % (2**31) + _DropoutFn._seed_counter[0]
+        return _DropoutFn.apply(x, p, seed)
+    return F.dropout(x, p, training)
+
+
+# --------------------------------------------------------------------------
+# Loss scaling utilities (K13, fp16 path)
+# --------------------------------------------------------------------------
+
+def unscale_and_check_finite(flats, inv_scale: float) -> bool:
+    """In-place multiply each flat grad buffer by inv_scale; returns True if
+    all values finite (fp16 dynamic loss scaling)."""
+    if flats and flats[0].is_cuda:
+        ext = load_kernels_ext()
+        found_inf = torch.zeros(1, dtype=torch.int32, device=flats[0].device)
+        for f in flats:
+            ext.unscale_check(f, float(inv_scale), found_inf)
+        return int(found_inf.item()) == 0
+    ok = True
+    for f in flats:
+        f.mul_(inv_scale)
+        ok = ok and bool(torch.isfinite(f).all())
+    return ok

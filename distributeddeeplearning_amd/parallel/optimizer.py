@@ -1,0 +1,156 @@
+"""DistributedOptimizer — hvd.DistributedOptimizer-shaped wrapper.
+
+Reference mechanism (Horovod ``horovod/torch/optimizer.py``): a hook on each
+parameter's grad accumulator launches an async allreduce during backward;
+``step()`` synchronises every handle then runs the inner optimizer.  Here the
+hook marks the param ready in its static bucket; a full bucket launches ONE
+allreduce of its persistent flat buffer (no negotiation round, no fusion
+memcpys — SURVEY.md §2.2 "Controller" row: static bucket order).
+
+Overlap: on GPU the allreduce runs on the comm backend's side HIP stream;
+``synchronize()`` makes the compute stream wait on comm events device-side,
+so the host never blocks between backward and step.
+"""
+from __future__ import annotations
+
+import os
+import time
+import warnings
+from typing import Optional
+
+import torch
+
+from . import api
+from .bucket import BucketManager
+
+STALL_WARN_SECONDS = float(os.environ.get("DDL_STALL_WARN_SECONDS", "60"))
+
+
+class DistributedOptimizer:
+    """Wraps a torch-style optimizer with bucketed gradient all-reduce.
+
+    Parameters mirror hvd.DistributedOptimizer: the wrapped ``optimizer``,
+    ``named_parameters`` (defining bucket membership and order), and
+    ``backward_passes_per_step`` for gradient accumulation.
+    """
+
+    def __init__(
+        self,
+        optimizer: torch.optim.Optimizer,
+        named_parameters=None,
+        backward_passes_per_step: int = 1,
+        bucket_bytes: Optional[int] = None,
+        average: bool = True,
+    ):
+        self.optimizer = optimizer
+        self.backward_passes_per_step = int(backward_passes_per_step)
+        self._passes = 0
+        self.average = average
+        self._timeline = None
+
+        if named_parameters is None:
+            named_parameters = [
+                (f"param.{gi}.{pi}", p)
+                for gi, g in enumerate(optimizer.param_groups)
+                for pi, p in enumerate(g["params"])
+            ]
+        else:
+            named_parameters = list(named_parameters)
+
+        self.buckets = BucketManager(named_parameters, bucket_bytes=bucket_bytes)
+        self._hooks = []
+        self._last_ready_time: Optional[float] = None
+        self._warned_stall = False
+        for b in self.buckets.buckets:
+            for p in b.params:
+                h = p.register_post_accumulate_grad_hook(self._make_hook())
+                self._hooks.append(h)
+
+    # -- hook & launch ------------------------------------------------------
+
+    def _make_hook(self):
+        def hook(p: torch.nn.Parameter):
+            self._last_ready_time = time.monotonic()
+            bucket = self.buckets.mark_ready(p)
+            if bucket is not None and self._should_reduce():
+                self._launch(bucket)
+
+        return hook
+
+    def _should_reduce(self) -> bool:
+        return (self._passes + 1) >= self.backward_passes_per_step
+
+    def _launch(self, bucket) -> None:
+        if api.size() > 1:
+            if self._timeline is not None:
+                self._timeline.mark(f"bucket{bucket.index}/launch")
+            bucket.handle = api.allreduce_async_(bucket.flat, average=self.average)
+
+    # -- hvd surface --------------------------------------------------------
+
+    def synchronize(self) -> None:
+        """Wait every in-flight bucket (device-side on GPU backends).
+
+        Also launches any bucket that never filled (a param that got no grad
+        this pass would otherwise deadlock other ranks — the stall-inspector
+        case).
+        """
+        for b in self.buckets.buckets:
+            if b.handle is None and self._should_reduce() and api.size() > 1:
+                if 0 < b.ready < len(b.params):
+                    warnings.warn(
+                        f"bucket {b.index}: only {b.ready}/{len(b.params)} grads "
+                        "produced this step; reducing it anyway (did part of the "
+                        "model not participate in the loss?)"
+                    )
+                    self._launch(b)
+            if b.handle is not None:
+                b.handle.wait()
+                b.handle = None
+
+    def step(self, closure=None):
+        self._passes += 1
+        if self._passes < self.backward_passes_per_step:
+            # accumulation step: no reduce, no inner step
+            self.buckets.reset()
+            return None
+        self.synchronize()
+        loss = self.optimizer.step(closure)
+        self._passes = 0
+        return loss
+
+    def zero_grad(self, set_to_none: bool = False):
+        # Grads are views into flat buckets; zero the buckets and keep views.
+        self.buckets.zero_grads()
+        self.buckets.reset()
+        if not self.buckets.grad_views_valid():
+            self.buckets.rebind_grads()
+
+    # Pass-through of the torch optimizer surface ---------------------------
+    @property
+    def param_groups(self):
+        return self.optimizer.param_groups
+
+    @property
+    def state(self):
+        return self.optimizer.state
+
+    def state_dict(self):
+        return self.optimizer.state_dict()
+
+    def load_state_dict(self, sd):
+        return self.optimizer.load_state_dict(sd)
+
+    def add_param_group(self, g):
+        raise NotImplementedError(
+            "add_param_group after DistributedOptimizer construction is not "
+            "supported: bucket membership is static"
+        )
+
+    def set_timeline(self, timeline):
+        self._timeline = timeline
+
+    def __repr__(self):
+        nb = len(self.buckets.buckets)
+        mb = self.buckets.total_bytes() / (1 << 20)
+        return f"DistributedOptimizer({self.optimizer.__class__.__name__}, buckets={nb}, {mb:.1f} MB)"

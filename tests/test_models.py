@@ -1,0 +1,70 @@
+"""Model construction, shapes, parameter counts and torchvision-compatible
+state-dict naming (checkpoint layout contract)."""
+import pytest
+import torch
+
+from distributeddeeplearning_amd import models
+
+
+@pytest.mark.parametrize("name,params_m", [
+    ("resnet18", 11.69), ("resnet50", 25.56), ("resnet152", 60.19),
+    ("vgg16", 138.36),
+])
+def test_param_counts(name, params_m):
+    m = models.build_model(name)
+    n = sum(p.numel() for p in m.parameters()) / 1e6
+    assert abs(n - params_m) < 0.05, f"{name}: {n:.2f}M params"
+
+
+@pytest.mark.parametrize("name", ["resnet18", "resnet50"])
+def test_forward_shape(name):
+    m = models.build_model(name, num_classes=10).eval()
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        y = m(x)
+    assert y.shape == (2, 10)
+
+
+def test_resnet50_state_dict_names():
+    sd = models.resnet50().state_dict()
+    expected = [
+        "conv1.weight", "bn1.weight", "bn1.bias", "bn1.running_mean",
+        "bn1.running_var", "bn1.num_batches_tracked",
+        "layer1.0.conv1.weight", "layer1.0.bn1.weight",
+        "layer1.0.downsample.0.weight", "layer1.0.downsample.1.running_var",
+        "layer4.2.bn3.bias", "fc.weight", "fc.bias",
+    ]
+    for k in expected:
+        assert k in sd, f"missing {k}"
+    # shapes of a few torchvision-known entries
+    assert sd["conv1.weight"].shape == (64, 3, 7, 7)
+    assert sd["layer1.0.conv2.weight"].shape == (64, 64, 3, 3)
+    assert sd["layer4.2.conv3.weight"].shape == (2048, 512, 1, 1)
+    assert sd["fc.weight"].shape == (1000, 2048)
+
+
+def test_vgg16_state_dict_names():
+    sd = models.vgg16().state_dict()
+    for k in ["features.0.weight", "features.0.bias", "features.2.weight",
+              "features.5.weight", "features.28.weight",
+              "classifier.0.weight", "classifier.3.weight",
+              "classifier.6.weight"]:
+        assert k in sd, f"missing {k}"
+    assert sd["features.0.weight"].shape == (64, 3, 3, 3)
+    assert sd["classifier.0.weight"].shape == (4096, 25088)
+
+
+def test_resnet_backward_cpu():
+    m = models.resnet18(num_classes=10)
+    x = torch.randn(2, 3, 32, 32)
+    y = torch.randint(0, 10, (2,))
+    from distributeddeeplearning_amd.ops import functional as DF
+    loss = DF.softmax_cross_entropy(m(x), y)
+    loss.backward()
+    assert all(p.grad is not None for p in m.parameters())
+    assert torch.isfinite(loss)
+
+
+def test_resnet152_structure():
+    m = models.resnet152()
+    assert len(m.layer3) == 36 and len(m.layer2) == 8
