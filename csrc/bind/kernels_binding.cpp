@@ -1,0 +1,348 @@
+// Torch extension binding for the HIP kernel library (csrc/kernels).
+// Host-side only: validates layouts/dtypes, allocates outputs (torch caching
+// allocator), forwards raw pointers + the current HIP stream to the C ABI.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <c10/util/Optional.h>
+
+#include "kernels/api.h"
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+int dt_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return 0;
+    case at::kBFloat16: return 1;
+    case at::kHalf: return 2;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+at::Tensor empty_nhwc(int64_t n, int64_t c, int64_t h, int64_t w,
+                      const at::TensorOptions& opts) {
+  return at::empty({n, c, h, w}, opts.memory_format(at::MemoryFormat::ChannelsLast));
+}
+
+void check_cl(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous(at::MemoryFormat::ChannelsLast) ||
+                  (t.dim() == 4 && (t.size(1) == 1 || (t.size(2) == 1 && t.size(3) == 1))) ||
+                  t.is_contiguous(),
+              name, " must be channels_last contiguous");
+}
+
+struct Conv {
+  int N, H, W, C, K, R, S, P, Q;
+};
+
+Conv conv_geom(const at::Tensor& x, const at::Tensor& w, int sh, int sw,
+               int ph, int pw) {
+  Conv c;
+  c.N = x.size(0); c.C = x.size(1); c.H = x.size(2); c.W = x.size(3);
+  c.K = w.size(0); c.R = w.size(2); c.S = w.size(3);
+  TORCH_CHECK(w.size(1) == c.C, "conv channel mismatch");
+  c.P = (c.H + 2 * ph - c.R) / sh + 1;
+  c.Q = (c.W + 2 * pw - c.S) / sw + 1;
+  return c;
+}
+
+// ---- conv ------------------------------------------------------------------
+
+at::Tensor conv_fwd(at::Tensor x, at::Tensor w,
+                    c10::optional<at::Tensor> bias, int64_t sh, int64_t sw,
+                    int64_t ph, int64_t pw, bool relu) {
+  check_cl(x, "x"); check_cl(w, "w");
+  TORCH_CHECK(x.scalar_type() == w.scalar_type());
+  TORCH_CHECK(x.scalar_type() != at::kFloat,
+              "GPU conv supports bf16/fp16 (fp32 path not built; use bf16)");
+  Conv g = conv_geom(x, w, sh, sw, ph, pw);
+  auto y = empty_nhwc(g.N, g.K, g.P, g.Q, x.options());
+  const void* bptr = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == x.scalar_type());
+    bptr = bias->data_ptr();
+  }
+  ddl_conv_fwd(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr(), g.N, g.H, g.W,
+               g.C, g.K, g.R, g.S, g.P, g.Q, sh, sw, ph, pw, relu,
+               dt_code(x), cur_stream());
+  return y;
+}
+
+at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
+                      std::vector<int64_t> x_shape, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw) {
+  check_cl(dy, "dy"); check_cl(w, "w");
+  int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
+  int K = w.size(0), R = w.size(2), S = w.size(3);
+  int P = dy.size(2), Q = dy.size(3);
+  TORCH_CHECK(C % 8 == 0, "dgrad requires C % 8 == 0 (skip first conv)");
+  // build CRSK-transposed weight for contiguous dgrad B-tiles
+  auto w_t = at::empty({(int64_t)C * R * S * K}, w.options());
+  ddl_transpose_krsc_crsk(w.data_ptr(), w_t.data_ptr(), K, R * S, C,
+                          dt_code(w), cur_stream());
+  auto dx = empty_nhwc(N, C, H, W, dy.options());
+  ddl_conv_dgrad(dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(), N, H, W, C, K,
+                 R, S, P, Q, sh, sw, ph, pw, dt_code(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
+                      std::vector<int64_t> w_shape, int64_t sh, int64_t sw,
+                      int64_t ph, int64_t pw) {
+  check_cl(dy, "dy"); check_cl(x, "x");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int K = w_shape[0], R = w_shape[2], S = w_shape[3];
+  int P = dy.size(2), Q = dy.size(3);
+  auto ws = at::zeros({(int64_t)K * R * S * C},
+                      x.options().dtype(at::kFloat));
+  ddl_conv_wgrad(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(), N, H, W,
+                 C, K, R, S, P, Q, sh, sw, ph, pw, dt_code(x), cur_stream());
+  auto dw = empty_nhwc(K, C, R, S, x.options());
+  ddl_cast_f32_to(ws.data_ptr<float>(), dw.data_ptr(), (int64_t)K * R * S * C,
+                  dt_code(x), cur_stream());
+  return dw;
+}
+
+// ---- batchnorm -------------------------------------------------------------
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
+    at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor rmean,
+    at::Tensor rvar, double momentum, double eps, bool relu,
+    c10::optional<at::Tensor> residual) {
+  check_cl(x, "x");
+  int64_t C = x.size(1);
+  int64_t M = x.numel() / C;
+  TORCH_CHECK(C % 8 == 0, "BN requires C % 8 == 0");
+  TORCH_CHECK(rmean.scalar_type() == at::kFloat && rvar.scalar_type() == at::kFloat,
+              "running stats must stay fp32");
+  TORCH_CHECK(weight.scalar_type() == x.scalar_type());
+  auto f32 = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2 * C}, f32);
+  auto mean = at::empty({C}, f32);
+  auto rstd = at::empty({C}, f32);
+  auto y = at::empty_like(x);
+  auto s = cur_stream();
+  int dt = dt_code(x);
+  ddl_bn_stats(x.data_ptr(), M, C, dt, sums.data_ptr<float>(), s);
+  ddl_bn_finalize(sums.data_ptr<float>(), C, M, momentum, eps,
+                  rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), 1, s);
+  const void* res = residual.has_value() && residual->defined()
+                        ? residual->data_ptr() : nullptr;
+  ddl_bn_apply(x.data_ptr(), y.data_ptr(), res, mean.data_ptr<float>(),
+               rstd.data_ptr<float>(), weight.data_ptr(), bias.data_ptr(), M,
+               C, relu, dt, s);
+  return {y, mean, rstd};
+}
+
+at::Tensor bn_relu_infer(at::Tensor x, at::Tensor weight, at::Tensor bias,
+                         at::Tensor rmean, at::Tensor rvar, double eps,
+                         bool relu, c10::optional<at::Tensor> residual) {
+  check_cl(x, "x");
+  int64_t C = x.size(1);
+  int64_t M = x.numel() / C;
+  auto y = at::empty_like(x);
+  const void* res = residual.has_value() && residual->defined()
+                        ? residual->data_ptr() : nullptr;
+  ddl_bn_infer(x.data_ptr(), y.data_ptr(), res, rmean.data_ptr<float>(),
+               rvar.data_ptr<float>(), eps, weight.data_ptr(),
+               bias.data_ptr(), M, C, relu, dt_code(x), cur_stream());
+  return y;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
+    at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor mean,
+    at::Tensor rstd, at::Tensor y, bool relu, bool has_residual) {
+  check_cl(dy, "dy"); check_cl(x, "x");
+  int64_t C = x.size(1);
+  int64_t M = x.numel() / C;
+  auto s = cur_stream();
+  int dt = dt_code(x);
+  auto sums = at::zeros({2 * C}, x.options().dtype(at::kFloat));
+  ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu,
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(), M, C, dt,
+                   sums.data_ptr<float>(), s);
+  auto dx = at::empty_like(x);
+  at::Tensor dres;
+  void* dres_p = nullptr;
+  if (has_residual) {
+    dres = at::empty_like(x);
+    dres_p = dres.data_ptr();
+  }
+  ddl_bn_bwd_apply(dy.data_ptr(), x.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                   dres_p, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   weight.data_ptr(), sums.data_ptr<float>(), M, C, relu, dt,
+                   s);
+  auto dw = at::empty_like(weight);
+  auto db = at::empty_like(weight);
+  ddl_bn_grads_from_sums(sums.data_ptr<float>(), dw.data_ptr(), db.data_ptr(),
+                         C, dt_code(weight), s);
+  return {dx, dw, db, dres};
+}
+
+// ---- pooling ---------------------------------------------------------------
+
+std::tuple<at::Tensor, at::Tensor> maxpool_fwd(at::Tensor x, int64_t kh,
+                                               int64_t kw, int64_t sh,
+                                               int64_t sw, int64_t ph,
+                                               int64_t pw) {
+  check_cl(x, "x");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int P = (H + 2 * ph - kh) / sh + 1;
+  int Q = (W + 2 * pw - kw) / sw + 1;
+  TORCH_CHECK(C % 8 == 0);
+  auto y = empty_nhwc(N, C, P, Q, x.options());
+  auto idx = empty_nhwc(N, C, P, Q, x.options().dtype(at::kByte));
+  ddl_maxpool_fwd(x.data_ptr(), y.data_ptr(), (uint8_t*)idx.data_ptr(), N, H,
+                  W, C, P, Q, kh, kw, sh, sw, ph, pw, dt_code(x),
+                  cur_stream());
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx,
+                       std::vector<int64_t> x_shape, int64_t kh, int64_t kw,
+                       int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  check_cl(dy, "dy");
+  int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
+  int P = dy.size(2), Q = dy.size(3);
+  auto dx = empty_nhwc(N, C, H, W, dy.options());
+  ddl_maxpool_bwd(dy.data_ptr(), (const uint8_t*)idx.data_ptr(),
+                  dx.data_ptr(), N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw,
+                  dt_code(dy), cur_stream());
+  return dx;
+}
+
+at::Tensor gap_fwd(at::Tensor x) {
+  check_cl(x, "x");
+  int N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(C % 8 == 0);
+  auto y = at::empty({N, C}, x.options());
+  ddl_gap_fwd(x.data_ptr(), y.data_ptr(), N, HW, C, dt_code(x), cur_stream());
+  return y;
+}
+
+at::Tensor gap_bwd(at::Tensor dy, std::vector<int64_t> x_shape) {
+  int N = x_shape[0], C = x_shape[1], HW = x_shape[2] * x_shape[3];
+  auto dx = empty_nhwc(N, C, x_shape[2], x_shape[3], dy.options());
+  ddl_gap_bwd(dy.contiguous().data_ptr(), dx.data_ptr(), N, HW, C,
+              dt_code(dy), cur_stream());
+  return dx;
+}
+
+// ---- softmax-xent ----------------------------------------------------------
+
+std::tuple<at::Tensor, at::Tensor> softmax_xent_fwd(at::Tensor logits,
+                                                    at::Tensor target,
+                                                    double ls) {
+  TORCH_CHECK(logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(target.scalar_type() == at::kLong);
+  int N = logits.size(0), C = logits.size(1);
+  auto f32 = logits.options().dtype(at::kFloat);
+  auto losses = at::empty({N}, f32);
+  auto probs = at::empty({N, C}, f32);
+  ddl_softmax_xent_fwd(logits.data_ptr(), target.data_ptr<int64_t>(),
+                       losses.data_ptr<float>(), probs.data_ptr<float>(), N,
+                       C, dt_code(logits), ls, cur_stream());
+  return {losses, probs};
+}
+
+at::Tensor softmax_xent_bwd(at::Tensor probs, at::Tensor target, double ls,
+                            at::Tensor dloss, int64_t out_dtype) {
+  int N = probs.size(0), C = probs.size(1);
+  auto dtype = out_dtype == 0 ? at::kFloat
+             : out_dtype == 1 ? at::kBFloat16 : at::kHalf;
+  auto dlogits = at::empty({N, C}, probs.options().dtype(dtype));
+  TORCH_CHECK(dloss.scalar_type() == at::kFloat && dloss.is_cuda());
+  ddl_softmax_xent_bwd(probs.data_ptr<float>(), target.data_ptr<int64_t>(),
+                       dlogits.data_ptr(), N, C, (int)out_dtype, ls,
+                       dloss.data_ptr<float>(), 1.0f / N, cur_stream());
+  return dlogits;
+}
+
+// ---- elementwise / dropout / scaling ---------------------------------------
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y) {
+  auto out = at::empty_like(dy);
+  ddl_relu_bwd(dy.data_ptr(), y.data_ptr(), out.data_ptr(), dy.numel(),
+               dt_code(dy), cur_stream());
+  return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> dropout_fwd(at::Tensor x, double p,
+                                               int64_t seed) {
+  auto y = at::empty_like(x);
+  auto mask = at::empty({x.numel()}, x.options().dtype(at::kByte));
+  ddl_dropout_fwd(x.data_ptr(), y.data_ptr(), (uint8_t*)mask.data_ptr(),
+                  x.numel(), dt_code(x), p, (uint64_t)seed, cur_stream());
+  return {y, mask};
+}
+
+at::Tensor dropout_bwd(at::Tensor dy, at::Tensor mask, double p) {
+  auto dx = at::empty_like(dy);
+  ddl_dropout_bwd(dy.data_ptr(), dx.data_ptr(),
+                  (const uint8_t*)mask.data_ptr(), dy.numel(), dt_code(dy), p,
+                  cur_stream());
+  return dx;
+}
+
+void check_finite(at::Tensor t, at::Tensor found_inf) {
+  ddl_check_finite(t.data_ptr(), t.numel(), dt_code(t),
+                   found_inf.data_ptr<int>(), cur_stream());
+}
+
+void unscale_check(at::Tensor t, double inv_scale, at::Tensor found_inf) {
+  ddl_unscale_check(t.data_ptr(), t.numel(), dt_code(t), inv_scale,
+                    found_inf.data_ptr<int>(), cur_stream());
+}
+
+// ---- fused optimizer -------------------------------------------------------
+
+void fused_sgd(at::Tensor table, c10::optional<at::Tensor> norms,
+               int64_t dtype, double lr, double momentum, double wd,
+               bool nesterov, double lars_eta, double lars_eps,
+               double inv_scale) {
+  int nrows = table.size(0);
+  const float* nptr = nullptr;
+  int ntensors = 0;
+  if (norms.has_value() && norms->defined()) {
+    nptr = norms->data_ptr<float>();
+    ntensors = norms->size(1);
+  }
+  ddl_fused_sgd(table.data_ptr<int64_t>(), nrows, nptr, ntensors, dtype, lr,
+                momentum, wd, nesterov, lars_eta, lars_eps, inv_scale,
+                cur_stream());
+}
+
+void lars_norms(at::Tensor table, at::Tensor norms, int64_t dtype, double wd,
+                double inv_scale) {
+  ddl_lars_norms(table.data_ptr<int64_t>(), table.size(0),
+                 norms.data_ptr<float>(), norms.size(1), dtype, wd, inv_scale,
+                 cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv_fwd", &conv_fwd);
+  m.def("conv_dgrad", &conv_dgrad);
+  m.def("conv_wgrad", &conv_wgrad);
+  m.def("bn_relu_fwd", &bn_relu_fwd);
+  m.def("bn_relu_infer", &bn_relu_infer);
+  m.def("bn_relu_bwd", &bn_relu_bwd);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("gap_fwd", &gap_fwd);
+  m.def("gap_bwd", &gap_bwd);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+  m.def("check_finite", &check_finite);
+  m.def("unscale_check", &unscale_check);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("lars_norms", &lars_norms);
+}

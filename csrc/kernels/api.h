@@ -1,0 +1,108 @@
+// C ABI between the HIP kernel library (hipcc-compiled, torch-free) and the
+// torch extension binding (csrc/bind/kernels_binding.cpp).
+// dtype codes: 0 = f32, 1 = bf16, 2 = f16.  All activation tensors are NHWC
+// physical (torch channels_last); M = N*H*W rows over C channels.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+extern "C" {
+
+// ---- elementwise / utility ------------------------------------------------
+void ddl_relu_bwd(const void* dy, const void* y, void* out, int64_t n,
+                  int dtype, hipStream_t s);
+void ddl_check_finite(const void* g, int64_t n, int dtype, int* found_inf,
+                      hipStream_t s);
+void ddl_unscale_check(void* g, int64_t n, int dtype, float inv_scale,
+                       int* found_inf, hipStream_t s);
+void ddl_dropout_fwd(const void* x, void* y, uint8_t* mask, int64_t n,
+                     int dtype, float p, uint64_t seed, hipStream_t s);
+void ddl_dropout_bwd(const void* dy, void* dx, const uint8_t* mask, int64_t n,
+                     int dtype, float p, hipStream_t s);
+void ddl_cast_f32_to(const float* src, void* dst, int64_t n, int dtype,
+                     hipStream_t s);
+
+// ---- fused optimizer (K10/K11) -------------------------------------------
+// table: rows of [param_ptr, grad_ptr, master_ptr, mom_ptr, numel, use_lars,
+// tensor_idx] (int64), one row per block.  norms: [2, ntensors] f32
+// (0: sum w^2 of master, 1: sum g'^2 where g' = g*inv_scale + wd*w).
+void ddl_fused_sgd(const int64_t* table, int nrows, const float* norms,
+                   int ntensors, int dtype, float lr, float momentum,
+                   float wd, int nesterov, float lars_eta, float lars_eps,
+                   float inv_scale, hipStream_t s);
+void ddl_lars_norms(const int64_t* table, int nrows, float* norms,
+                    int ntensors, int dtype, float wd, float inv_scale,
+                    hipStream_t s);
+
+// ---- fused BatchNorm + residual + ReLU (K4/K5/K14) ------------------------
+// sums: f32 [2*C] workspace (pre-zeroed): 0..C-1 = sum x, C..2C-1 = sum x^2.
+void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
+                  hipStream_t s);
+// finalize: mean/rstd from sums; optionally update running stats in-place.
+void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
+                     float eps, float* running_mean, float* running_var,
+                     float* mean_out, float* rstd_out, int update_running,
+                     hipStream_t s);
+// y = relu?( (x-mean)*rstd*w + b + residual? )   (w/b in x's dtype)
+void ddl_bn_apply(const void* x, void* y, const void* residual,
+                  const float* mean, const float* rstd, const void* weight,
+                  const void* bias, int64_t M, int C, int relu, int dtype,
+                  hipStream_t s);
+// inference: scale/shift from running stats, single pass
+void ddl_bn_infer(const void* x, void* y, const void* residual,
+                  const float* running_mean, const float* running_var,
+                  float eps, const void* weight, const void* bias, int64_t M,
+                  int C, int relu, int dtype, hipStream_t s);
+// backward stats: sums[0..C-1] = sum(dy_m * xhat), sums[C..] = sum(dy_m)
+// where dy_m = dy * (y>0) if relu else dy.  Pre-zeroed.
+void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
+                      const float* mean, const float* rstd, int64_t M, int C,
+                      int dtype, float* sums, hipStream_t s);
+// dx (+ optional dres copy of masked dy), dweight/dbias cast from sums.
+void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y,
+                      void* dx, void* dres, const float* mean,
+                      const float* rstd, const void* weight, const float* sums,
+                      int64_t M, int C, int relu, int dtype, hipStream_t s);
+void ddl_bn_grads_from_sums(const float* sums, void* dweight, void* dbias,
+                            int C, int dtype, hipStream_t s);
+
+// ---- pooling (K6/K7) ------------------------------------------------------
+void ddl_maxpool_fwd(const void* x, void* y, uint8_t* idx, int N, int H,
+                     int W, int C, int P, int Q, int kh, int kw, int sh,
+                     int sw, int ph, int pw, int dtype, hipStream_t s);
+void ddl_maxpool_bwd(const void* dy, const uint8_t* idx, void* dx, int N,
+                     int H, int W, int C, int P, int Q, int kh, int kw,
+                     int sh, int sw, int ph, int pw, int dtype, hipStream_t s);
+void ddl_gap_fwd(const void* x, void* y, int N, int HW, int C, int dtype,
+                 hipStream_t s);
+void ddl_gap_bwd(const void* dy, void* dx, int N, int HW, int C, int dtype,
+                 hipStream_t s);
+
+// ---- fused softmax + cross-entropy (K9) -----------------------------------
+void ddl_softmax_xent_fwd(const void* logits, const int64_t* target,
+                          float* losses, float* probs, int N, int C,
+                          int dtype, float label_smoothing, hipStream_t s);
+void ddl_softmax_xent_bwd(const float* probs, const int64_t* target,
+                          void* dlogits, int N, int C, int dtype,
+                          float label_smoothing, const float* scale_ptr,
+                          float scale_const, hipStream_t s);
+
+// ---- implicit-GEMM conv (K1/K2/K3), NHWC, MFMA ----------------------------
+// x [N,H,W,C], w [K,R,S,C] (KRSC = channels_last), y [N,P,Q,K].
+void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
+                  int N, int H, int W, int C, int K, int R, int S, int P,
+                  int Q, int sh, int sw, int ph, int pw, int relu, int dtype,
+                  hipStream_t s);
+// dgrad consumes w_t [C,R,S,K] (CRSK) built by ddl_transpose_krsc_crsk.
+void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
+                    int W, int C, int K, int R, int S, int P, int Q, int sh,
+                    int sw, int ph, int pw, int dtype, hipStream_t s);
+// wgrad accumulates fp32 into dw_ws [K,R,S,C] (pre-zeroed).
+void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, int N,
+                    int H, int W, int C, int K, int R, int S, int P, int Q,
+                    int sh, int sw, int ph, int pw, int dtype, hipStream_t s);
+// [K, RS, C] -> [C, RS, K] element transpose (K12 utility).
+void ddl_transpose_krsc_crsk(const void* w, void* w_t, int K, int RS, int C,
+                             int dtype, hipStream_t s);
+
+}  // extern "C"

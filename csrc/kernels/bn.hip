@@ -1,0 +1,359 @@
+// K4/K5/K14: fused BatchNorm + residual-add + ReLU over NHWC (M rows x C).
+// Training forward = 3 launches: per-channel {sum, sumsq} partials with
+// atomic accumulation (stats), a tiny finalize (mean/rstd + running-stat
+// update, fp32), and one fused normalize+add+relu pass.  The reference
+// issues separate cuDNN BN, add and ReLU kernels (2-3 full activation
+// round-trips more traffic — SURVEY.md K4/K14).
+// Constraint: C % 8 == 0 (vectorized channel groups); every BN in
+// ResNet/VGG has C in {64..2048}.
+#include "common.h"
+#include "api.h"
+
+namespace {
+
+constexpr int ROWS = 32;     // rows per block pass
+constexpr int CG = 8;        // channel strip = CG vec8-groups = 64 channels
+
+// ---- stats: sums[c] += sum_m x, sums[C+c] += sum_m x^2 --------------------
+template <typename T>
+__global__ void bn_stats_k(const T* __restrict__ x, int64_t M, int C,
+                           float* __restrict__ sums) {
+  // block: 256 threads = CG vec-groups x ROWS rows; grid.x = C/64 strips,
+  // grid.y = row chunks.
+  __shared__ float red[2][ROWS][CG * 8];
+  const int vg = threadIdx.x % CG;          // which vec8 group in the strip
+  const int trow = threadIdx.x / CG;        // 0..31
+  const int c0 = blockIdx.x * CG * 8 + vg * 8;
+  if (c0 >= C) return;
+  float s[8] = {0}, q[8] = {0};
+  for (int64_t m = (int64_t)blockIdx.y * ROWS + trow; m < M;
+       m += (int64_t)gridDim.y * ROWS) {
+    Vec8<T> v = load8(x + m * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(v.v[j]);
+      s[j] += f;
+      q[j] += f * f;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][trow][vg * 8 + j] = s[j];
+    red[1][trow][vg * 8 + j] = q[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int step = ROWS / 2; step > 0; step >>= 1) {
+    if (trow < step) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        red[0][trow][vg * 8 + j] += red[0][trow + step][vg * 8 + j];
+        red[1][trow][vg * 8 + j] += red[1][trow + step][vg * 8 + j];
+      }
+    }
+    __syncthreads();
+  }
+  if (trow == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sums[c0 + j], red[0][0][vg * 8 + j]);
+      atomicAdd(&sums[C + c0 + j], red[1][0][vg * 8 + j]);
+    }
+  }
+}
+
+__global__ void bn_finalize_k(const float* __restrict__ sums, int C,
+                              float inv_M, float unbias, float momentum,
+                              float eps, float* running_mean,
+                              float* running_var, float* mean_out,
+                              float* rstd_out, int update_running) {
+  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
+       c += blockDim.x * gridDim.x) {
+    float mean = sums[c] * inv_M;
+    float var = fmaxf(sums[C + c] * inv_M - mean * mean, 0.f);
+    mean_out[c] = mean;
+    rstd_out[c] = rsqrtf(var + eps);
+    if (update_running) {
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * var * unbias;
+    }
+  }
+}
+
+// ---- apply: y = relu?((x-mean)*rstd*w + b + res?) -------------------------
+template <typename T, bool RELU, bool RES>
+__global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
+                           const T* __restrict__ res,
+                           const float* __restrict__ mean,
+                           const float* __restrict__ rstd,
+                           const T* __restrict__ weight,
+                           const T* __restrict__ bias, int64_t M, int C) {
+  const int ncg = C / 8;
+  const int64_t total = M * ncg;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int cg = (int)(i % ncg);
+    const int64_t m = i / ncg;
+    const int c0 = cg * 8;
+    Vec8<T> v = load8(x + m * C + c0);
+    Vec8<T> w8 = load8(weight + c0);
+    Vec8<T> b8 = load8(bias + c0);
+    Vec8<T> r8;
+    if (RES) r8 = load8(res + m * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (to_f32(v.v[j]) - mean[c0 + j]) * rstd[c0 + j];
+      f = f * to_f32(w8.v[j]) + to_f32(b8.v[j]);
+      if (RES) f += to_f32(r8.v[j]);
+      if (RELU) f = fmaxf(f, 0.f);
+      v.v[j] = from_f32<T>(f);
+    }
+    store8(y + m * C + c0, v);
+  }
+}
+
+// inference: same but scale/shift precomputed from running stats on the fly
+template <typename T, bool RELU, bool RES>
+__global__ void bn_infer_k(const T* __restrict__ x, T* __restrict__ y,
+                           const T* __restrict__ res,
+                           const float* __restrict__ rmean,
+                           const float* __restrict__ rvar, float eps,
+                           const T* __restrict__ weight,
+                           const T* __restrict__ bias, int64_t M, int C) {
+  const int ncg = C / 8;
+  const int64_t total = M * ncg;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int cg = (int)(i % ncg);
+    const int64_t m = i / ncg;
+    const int c0 = cg * 8;
+    Vec8<T> v = load8(x + m * C + c0);
+    Vec8<T> w8 = load8(weight + c0);
+    Vec8<T> b8 = load8(bias + c0);
+    Vec8<T> r8;
+    if (RES) r8 = load8(res + m * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float rs = rsqrtf(rvar[c0 + j] + eps);
+      float f = (to_f32(v.v[j]) - rmean[c0 + j]) * rs;
+      f = f * to_f32(w8.v[j]) + to_f32(b8.v[j]);
+      if (RES) f += to_f32(r8.v[j]);
+      if (RELU) f = fmaxf(f, 0.f);
+      v.v[j] = from_f32<T>(f);
+    }
+    store8(y + m * C + c0, v);
+  }
+}
+
+// ---- backward stats: sums[c] = sum(dy_m * xhat), sums[C+c] = sum(dy_m) ----
+template <typename T, bool RELU>
+__global__ void bn_bwd_stats_k(const T* __restrict__ dy,
+                               const T* __restrict__ x,
+                               const T* __restrict__ y,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ rstd, int64_t M,
+                               int C, float* __restrict__ sums) {
+  __shared__ float red[2][ROWS][CG * 8];
+  const int vg = threadIdx.x % CG;
+  const int trow = threadIdx.x / CG;
+  const int c0 = blockIdx.x * CG * 8 + vg * 8;
+  if (c0 >= C) return;
+  float s0[8] = {0}, s1[8] = {0};
+  for (int64_t m = (int64_t)blockIdx.y * ROWS + trow; m < M;
+       m += (int64_t)gridDim.y * ROWS) {
+    Vec8<T> vdy = load8(dy + m * C + c0);
+    Vec8<T> vx = load8(x + m * C + c0);
+    Vec8<T> vy;
+    if (RELU) vy = load8(y + m * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = to_f32(vdy.v[j]);
+      if (RELU) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
+      float xh = (to_f32(vx.v[j]) - mean[c0 + j]) * rstd[c0 + j];
+      s0[j] += g * xh;
+      s1[j] += g;
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][trow][vg * 8 + j] = s0[j];
+    red[1][trow][vg * 8 + j] = s1[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int step = ROWS / 2; step > 0; step >>= 1) {
+    if (trow < step) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        red[0][trow][vg * 8 + j] += red[0][trow + step][vg * 8 + j];
+        red[1][trow][vg * 8 + j] += red[1][trow + step][vg * 8 + j];
+      }
+    }
+    __syncthreads();
+  }
+  if (trow == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sums[c0 + j], red[0][0][vg * 8 + j]);
+      atomicAdd(&sums[C + c0 + j], red[1][0][vg * 8 + j]);
+    }
+  }
+}
+
+// ---- backward apply: dx = w*rstd*(dy_m - s1/M - xhat*s0/M); dres = dy_m ---
+template <typename T, bool RELU, bool RES>
+__global__ void bn_bwd_apply_k(const T* __restrict__ dy,
+                               const T* __restrict__ x,
+                               const T* __restrict__ y, T* __restrict__ dx,
+                               T* __restrict__ dres,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ rstd,
+                               const T* __restrict__ weight,
+                               const float* __restrict__ sums, int64_t M,
+                               int C) {
+  const int ncg = C / 8;
+  const int64_t total = M * ncg;
+  const float invM = 1.f / (float)M;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int cg = (int)(i % ncg);
+    const int64_t m = i / ncg;
+    const int c0 = cg * 8;
+    Vec8<T> vdy = load8(dy + m * C + c0);
+    Vec8<T> vx = load8(x + m * C + c0);
+    Vec8<T> w8 = load8(weight + c0);
+    Vec8<T> vy, out, rout;
+    if (RELU) vy = load8(y + m * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j;
+      float g = to_f32(vdy.v[j]);
+      if (RELU) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
+      if (RES) rout.v[j] = from_f32<T>(g);
+      float xh = (to_f32(vx.v[j]) - mean[c]) * rstd[c];
+      float t = g - sums[C + c] * invM - xh * sums[c] * invM;
+      out.v[j] = from_f32<T>(to_f32(w8.v[j]) * rstd[c] * t);
+    }
+    store8(dx + m * C + c0, out);
+    if (RES) store8(dres + m * C + c0, rout);
+  }
+}
+
+template <typename T>
+__global__ void bn_grads_k(const float* sums, T* dw, T* db, int C) {
+  for (int c = threadIdx.x + blockIdx.x * blockDim.x; c < C;
+       c += blockDim.x * gridDim.x) {
+    dw[c] = from_f32<T>(sums[c]);
+    db[c] = from_f32<T>(sums[C + c]);
+  }
+}
+
+inline dim3 stats_grid(int64_t M, int C) {
+  int strips = (C + 63) / 64;
+  int64_t want = 2048 / strips;
+  int64_t chunks = std::min<int64_t>(cdiv(M, ROWS), std::max<int64_t>(want, 1));
+  return dim3(strips, (unsigned)chunks);
+}
+
+inline int apply_grid(int64_t M, int C) {
+  return (int)std::min<int64_t>(cdiv(M * (C / 8), 256), 2048);
+}
+
+}  // namespace
+
+extern "C" {
+
+void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
+                  hipStream_t s) {
+  dim3 g = stats_grid(M, C);
+#define RUN(T) bn_stats_k<T><<<g, 256, 0, s>>>((const T*)x, M, C, sums)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
+                     float eps, float* running_mean, float* running_var,
+                     float* mean_out, float* rstd_out, int update_running,
+                     hipStream_t s) {
+  float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  bn_finalize_k<<<cdiv(C, 256), 256, 0, s>>>(
+      sums, C, 1.f / (float)M, unbias, momentum, eps, running_mean,
+      running_var, mean_out, rstd_out, update_running);
+}
+
+void ddl_bn_apply(const void* x, void* y, const void* residual,
+                  const float* mean, const float* rstd, const void* weight,
+                  const void* bias, int64_t M, int C, int relu, int dtype,
+                  hipStream_t s) {
+  int g = apply_grid(M, C);
+#define RUN(T, RELU, RES)                                                      \
+  bn_apply_k<T, RELU, RES><<<g, 256, 0, s>>>(                                  \
+      (const T*)x, (T*)y, (const T*)residual, mean, rstd, (const T*)weight,    \
+      (const T*)bias, M, C)
+#define SEL(T)                                                                 \
+  if (relu && residual) RUN(T, true, true);                                    \
+  else if (relu) RUN(T, true, false);                                          \
+  else if (residual) RUN(T, false, true);                                      \
+  else RUN(T, false, false)
+  switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
+#undef SEL
+#undef RUN
+}
+
+void ddl_bn_infer(const void* x, void* y, const void* residual,
+                  const float* running_mean, const float* running_var,
+                  float eps, const void* weight, const void* bias, int64_t M,
+                  int C, int relu, int dtype, hipStream_t s) {
+  int g = apply_grid(M, C);
+#define RUN(T, RELU, RES)                                                      \
+  bn_infer_k<T, RELU, RES><<<g, 256, 0, s>>>(                                  \
+      (const T*)x, (T*)y, (const T*)residual, running_mean, running_var, eps,  \
+      (const T*)weight, (const T*)bias, M, C)
+#define SEL(T)                                                                 \
+  if (relu && residual) RUN(T, true, true);                                    \
+  else if (relu) RUN(T, true, false);                                          \
+  else if (residual) RUN(T, false, true);                                      \
+  else RUN(T, false, false)
+  switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
+#undef SEL
+#undef RUN
+}
+
+void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
+                      const float* mean, const float* rstd, int64_t M, int C,
+                      int dtype, float* sums, hipStream_t s) {
+  dim3 g = stats_grid(M, C);
+#define RUN(T, RELU)                                                           \
+  bn_bwd_stats_k<T, RELU><<<g, 256, 0, s>>>(                                   \
+      (const T*)dy, (const T*)x, (const T*)y, mean, rstd, M, C, sums)
+#define SEL(T)                                                                 \
+  if (relu) RUN(T, true); else RUN(T, false)
+  switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
+#undef SEL
+#undef RUN
+}
+
+void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y, void* dx,
+                      void* dres, const float* mean, const float* rstd,
+                      const void* weight, const float* sums, int64_t M, int C,
+                      int relu, int dtype, hipStream_t s) {
+  int g = apply_grid(M, C);
+#define RUN(T, RELU, RES)                                                      \
+  bn_bwd_apply_k<T, RELU, RES><<<g, 256, 0, s>>>(                              \
+      (const T*)dy, (const T*)x, (const T*)y, (T*)dx, (T*)dres, mean, rstd,    \
+      (const T*)weight, sums, M, C)
+#define SEL(T)                                                                 \
+  if (relu && dres) RUN(T, true, true);                                        \
+  else if (relu) RUN(T, true, false);                                          \
+  else if (dres) RUN(T, false, true);                                          \
+  else RUN(T, false, false)
+  switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
+#undef SEL
+#undef RUN
+}
+
+void ddl_bn_grads_from_sums(const float* sums, void* dweight, void* dbias,
+                            int C, int dtype, hipStream_t s) {
+#define RUN(T) bn_grads_k<T><<<cdiv(C, 256), 256, 0, s>>>(sums, (T*)dweight, (T*)dbias, C)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+}  // extern "C"

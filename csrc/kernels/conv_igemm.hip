@@ -1,0 +1,255 @@
+// K1 (forward) and K2 (dgrad) implicit-GEMM NHWC convolution for gfx950.
+//
+// Forward:  y[m=(n,p,q)][k] = sum_{r,s,c} x[n, p*sh+r-ph, q*sw+s-pw, c] *
+//                                         w[k,r,s,c]            (+bias,relu)
+// Dgrad:    dx[m=(n,h,w)][c] = sum_{r,s,k} dy[n,(h+ph-r)/sh,(w+pw-s)/sw,k] *
+//                                          w[k,r,s,c]   (terms where /sh,/sw
+//                                          divide exactly)
+// Both are the same GEMM loop: per (r,s), per 64-wide reduce tile, stage an
+// im2col A-tile (hardware-free zero padding via validity masks) and a
+// B^T-layout weight tile into swizzled LDS, then 2x16 MFMA per wave.
+// Dgrad consumes the CRSK-transposed weight so its B-tile loads are
+// contiguous (built once per backward by ddl_transpose_krsc_crsk).
+//
+// MODE_FLAT handles C not divisible by 8 (conv1: C=3) by flattening the
+// reduce dim to R*S*C with per-element scalar gather.
+#include "conv_tile.h"
+#include "api.h"
+
+namespace {
+
+enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2 };
+
+template <typename T, int MODE>
+__global__ __launch_bounds__(256) void conv_igemm_k(
+    const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
+    const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
+    int Nc, int relu) {
+  __shared__ __align__(16) char smem[2 * TILE_BYTES];
+  char* ldsA = smem;
+  char* ldsB = smem + TILE_BYTES;
+
+  const int nt_n = (Nc + BN - 1) / BN;
+  const unsigned bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  const int64_t tile_m = (int64_t)(bid / nt_n) * BM;
+  const int tile_n = (int)(bid % nt_n) * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 1) * 64;  // wave row offset in tile
+  const int wc = (wid & 1) * 64;   // wave col offset
+  const int trow = threadIdx.x >> 3;  // staging row base (0..31)
+  const int tcol = threadIdx.x & 7;   // staging 16B unit (0..7)
+
+  f32x4_t acc[4][4] = {};
+
+  // -- per-thread A-row decode (4 staged rows) ------------------------------
+  int64_t a_row_off[4];  // offset of the (n,hi,wi) pixel * C, or -1 invalid
+  int a_n[4], a_i1[4], a_i2[4];  // decoded (n, p|h, q|w) per staged row
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int64_t m = tile_m + trow + 32 * i;
+    if (m < M) {
+      if (MODE == DGRAD) {
+        a_i2[i] = (int)(m % d.W);
+        a_i1[i] = (int)((m / d.W) % d.H);
+        a_n[i] = (int)(m / ((int64_t)d.W * d.H));
+      } else {
+        a_i2[i] = (int)(m % d.Q);
+        a_i1[i] = (int)((m / d.Q) % d.P);
+        a_n[i] = (int)(m / ((int64_t)d.Q * d.P));
+      }
+    } else {
+      a_n[i] = -1;
+    }
+  }
+
+  const int Cred = (MODE == DGRAD) ? d.K : d.C;      // reduce length per rs
+  const int64_t a_ld = (MODE == DGRAD) ? d.K : d.C;  // A row stride (els)
+  const int b_ld = (MODE == DGRAD) ? d.K : d.C;      // B rs-slice row stride
+
+  const int rs_total = (MODE == FWD_FLAT) ? 1 : d.R * d.S;
+  const int kred_total = (MODE == FWD_FLAT) ? d.R * d.S * d.C : Cred;
+
+  for (int rs = 0; rs < rs_total; ++rs) {
+    const int r = rs / d.S, s = rs % d.S;
+    // Per-rs A row pointers + validity
+    const T* a_ptr[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      a_ptr[i] = nullptr;
+      if (a_n[i] < 0) continue;
+      if (MODE == FWD) {
+        const int hi = a_i1[i] * d.sh + r - d.ph;
+        const int wi = a_i2[i] * d.sw + s - d.pw;
+        if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+          a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
+      } else if (MODE == DGRAD) {
+        const int hn = a_i1[i] + d.ph - r;
+        const int wn = a_i2[i] + d.pw - s;
+        if (hn >= 0 && wn >= 0 && hn % d.sh == 0 && wn % d.sw == 0) {
+          const int ho = hn / d.sh, wo = wn / d.sw;
+          if (ho < d.P && wo < d.Q)
+            a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
+        }
+      } else {  // FWD_FLAT: row pixel base only (per-element taps)
+        a_ptr[i] = Asrc;  // marker: row exists
+      }
+    }
+    // B row (output-col) pointers for this rs
+    // fwd: row = k_out -> w[((k*R + r)*S + s)*C + c]
+    // dgrad: row = c_in -> w_t[((c*R + r)*S + s)*K + k]
+    const T* b_ptr[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row_n = tile_n + trow + 32 * i;
+      b_ptr[i] = nullptr;
+      if (row_n < Nc) {
+        if (MODE == FWD_FLAT)
+          b_ptr[i] = Bsrc + (int64_t)row_n * kred_total;
+        else
+          b_ptr[i] = Bsrc + ((int64_t)row_n * d.R + r) * d.S * b_ld +
+                     (int64_t)s * b_ld;
+      }
+    }
+
+    for (int c0 = 0; c0 < kred_total; c0 += BK) {
+      // ---- stage A ----
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = trow + 32 * i;
+        const int cel = c0 + tcol * 8;
+        uint4 v = zero16();
+        if (MODE == FWD_FLAT) {
+          if (a_n[i] >= 0) {
+            uint16_t* e = reinterpret_cast<uint16_t*>(&v);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int kred = cel + j;
+              if (kred < kred_total) {
+                const int rr = kred / (d.S * d.C);
+                const int rem = kred % (d.S * d.C);
+                const int ss = rem / d.C;
+                const int cc = rem % d.C;
+                const int hi = a_i1[i] * d.sh + rr - d.ph;
+                const int wi = a_i2[i] * d.sw + ss - d.pw;
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W) {
+                  const T* p = Asrc +
+                      (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc;
+                  e[j] = *reinterpret_cast<const uint16_t*>(p);
+                }
+              }
+            }
+          }
+        } else {
+          if (a_ptr[i] != nullptr && cel + 8 <= Cred)
+            v = *reinterpret_cast<const uint4*>(a_ptr[i] + cel);
+        }
+        lds_write16(ldsA, row, tcol * 16, v);
+      }
+      // ---- stage B ----
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = trow + 32 * i;
+        const int cel = c0 + tcol * 8;
+        uint4 v = zero16();
+        if (b_ptr[i] != nullptr) {
+          if (MODE == FWD_FLAT) {
+            uint16_t* e = reinterpret_cast<uint16_t*>(&v);
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              if (cel + j < kred_total)
+                e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
+          } else if (cel + 8 <= Cred) {
+            v = *reinterpret_cast<const uint4*>(b_ptr[i] + cel);
+          }
+        }
+        lds_write16(ldsB, row, tcol * 16, v);
+      }
+      __syncthreads();
+
+      // ---- MFMA ----
+      const int frow = lane & 15;
+      const int fcolb = (lane >> 4) * 16;  // byte offset of this lane's 8 els
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        typename MfmaFrag<T>::type a[4], b[4];
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr) {
+          uint4 v = lds_read16(ldsA, wr + mr * 16 + frow, kk * 64 + fcolb);
+          a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr) {
+          uint4 v = lds_read16(ldsB, wc + nr * 16 + frow, kk * 64 + fcolb);
+          b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: D[m][n], m = (lane>>4)*4 + reg, n = lane&15 --------------
+  const int en = lane & 15;
+  const int em = (lane >> 4) * 4;
+#pragma unroll
+  for (int nr = 0; nr < 4; ++nr) {
+    const int n = tile_n + wc + nr * 16 + en;
+    if (n >= Nc) continue;
+    const float bv = (MODE != DGRAD && bias != nullptr) ? to_f32(bias[n]) : 0.f;
+#pragma unroll
+    for (int mr = 0; mr < 4; ++mr) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int64_t m = tile_m + wr + mr * 16 + em + reg;
+        if (m >= M) continue;
+        float f = acc[mr][nr][reg] + bv;
+        if (relu) f = fmaxf(f, 0.f);
+        out[m * Nc + n] = from_f32<T>(f);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
+                  int N, int H, int W, int C, int K, int R, int S, int P,
+                  int Q, int sh, int sw, int ph, int pw, int relu, int dtype,
+                  hipStream_t s) {
+  ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
+  const int64_t M = (int64_t)N * P * Q;
+  const int Nc = K;
+  const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+  const bool flat = (C % 8) != 0;
+#define RUN(T, MODE)                                                           \
+  conv_igemm_k<T, MODE><<<(unsigned)grid, 256, 0, s>>>(                        \
+      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu)
+  if (dtype == 1) { if (flat) RUN(bf16, FWD_FLAT); else RUN(bf16, FWD); }
+  else if (dtype == 2) { if (flat) RUN(f16, FWD_FLAT); else RUN(f16, FWD); }
+#undef RUN
+}
+
+void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
+                    int W, int C, int K, int R, int S, int P, int Q, int sh,
+                    int sw, int ph, int pw, int dtype, hipStream_t s) {
+  ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
+  const int64_t M = (int64_t)N * H * W;
+  const int Nc = C;
+  const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+#define RUN(T)                                                                 \
+  conv_igemm_k<T, DGRAD><<<(unsigned)grid, 256, 0, s>>>(                       \
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0)
+  if (dtype == 1) RUN(bf16);
+  else if (dtype == 2) RUN(f16);
+#undef RUN
+}
+
+}  // extern "C"

@@ -1,0 +1,65 @@
+// Shared MFMA tile machinery for the implicit-GEMM conv kernels (K1-K3).
+// gfx950: v_mfma_f32_16x16x32_{bf16,f16}, 128x128x64 block tile, 4 waves
+// (2x2), 64x64 per wave as 4x4 fragments of 16x16.
+//
+// Fragment lane mapping (CDNA4, K=32 forms):
+//   A[i][k]:  i = lane&15,          k = (lane>>4)*8 + j   (j = 0..7)
+//   B[k][n]:  n = lane&15,          k = (lane>>4)*8 + j
+//   D[m][n]:  m = (lane>>4)*4 + r,  n = lane&15           (r = reg 0..3)
+//
+// LDS tiles are [128 rows][64 cols] elements (128 B rows) holding the
+// GEMM-transposed operand layouts (row = output row / output col, col =
+// reduce dim), XOR-swizzled with ((row&15)<<4) so the 16-lane ds_read_b128
+// groups (rows distinct mod 16) are bank-conflict-free (guide §6 G4 / T2).
+#pragma once
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+template <typename T> struct MfmaFrag;
+template <> struct MfmaFrag<bf16> { using type = bf16x8_t; };
+template <> struct MfmaFrag<f16> { using type = f16x8_t; };
+
+template <typename T>
+__device__ __forceinline__ f32x4_t mfma_16x16x32(typename MfmaFrag<T>::type a,
+                                                 typename MfmaFrag<T>::type b,
+                                                 f32x4_t c);
+template <>
+__device__ __forceinline__ f32x4_t mfma_16x16x32<bf16>(bf16x8_t a, bf16x8_t b,
+                                                       f32x4_t c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+template <>
+__device__ __forceinline__ f32x4_t mfma_16x16x32<f16>(f16x8_t a, f16x8_t b,
+                                                      f32x4_t c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+}
+
+// Tile geometry
+constexpr int BM = 128;   // output-row tile
+constexpr int BN = 128;   // output-col tile
+constexpr int BK = 64;    // reduce tile
+constexpr int TILE_BYTES = BM * BK * 2;  // 16 KiB per operand tile
+
+__device__ __forceinline__ int swz_off(int row, int colbyte) {
+  return ((row << 7) + colbyte) ^ ((row & 15) << 4);
+}
+
+__device__ __forceinline__ void lds_write16(char* lds, int row, int colbyte,
+                                            uint4 v) {
+  *reinterpret_cast<uint4*>(lds + swz_off(row, colbyte)) = v;
+}
+
+__device__ __forceinline__ uint4 lds_read16(const char* lds, int row,
+                                            int colbyte) {
+  return *reinterpret_cast<const uint4*>(lds + swz_off(row, colbyte));
+}
+
+__device__ __forceinline__ uint4 zero16() { return uint4{0, 0, 0, 0}; }
+
+// Decode helpers ------------------------------------------------------------
+struct ConvDims {
+  int N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw;
+};

@@ -1,0 +1,181 @@
+// K3: wgrad implicit GEMM — dw[k,r,s,c] = sum_m dy[m,k] * im2col_x[m,(r,s,c)].
+// Per (r,s) and m-chunk: GEMM tile dw[128k][128c], reduce over the chunk's
+// rows 64 at a time.  Both operands need the reduce dim (m) innermost in the
+// fragment, so dy and x tiles are TRANSPOSE-staGED into LDS ([k][m] / [c][m])
+// with scalar ds_write_b16 (v1; tr_b16 reads are the planned upgrade), then
+// the MFMA loop reads b128 like the forward kernel.  Chunks accumulate into
+// an fp32 workspace with atomicAdd (split-K over m; order nondeterministic
+// in the low bits, as cuDNN's default wgrad is).
+#include "conv_tile.h"
+#include "api.h"
+
+namespace {
+
+enum WMode { TILED = 0, FLAT = 1 };
+
+template <typename T, int MODE>
+__global__ __launch_bounds__(256) void conv_wgrad_k(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows) {
+  __shared__ __align__(16) char smem[2 * TILE_BYTES];
+  char* ldsA = smem;                 // dy^T tile: [128 k][64 m]
+  char* ldsB = smem + TILE_BYTES;    // x^T tile:  [128 c|kred][64 m]
+
+  const int kred_total = (MODE == FLAT) ? d.R * d.S * d.C : d.C;
+  const int ntc = (kred_total + BN - 1) / BN;
+  const int tile_k = (int)(blockIdx.x / ntc) * BM;
+  const int tile_c = (int)(blockIdx.x % ntc) * BN;
+  const int rs = (MODE == FLAT) ? 0 : blockIdx.y;
+  const int r = rs / d.S, s = rs % d.S;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = (wid >> 1) * 64;
+  const int wc = (wid & 1) * 64;
+  // staging assignment: unit u (0..15) covers 16B of a row of 128 2B els,
+  // m_local = t/16 (0..15) + 16*pass (4 passes -> 64 m)
+  const int su = threadIdx.x & 15;
+  const int sm = threadIdx.x >> 4;
+
+  f32x4_t acc[4][4] = {};
+
+  const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
+  const int64_t m_end = min(M, m_begin + chunk_rows);
+
+  for (int64_t m0 = m_begin; m0 < m_end; m0 += BK) {
+    // ---- stage dy^T: load dy[m][tile_k + u*8 ..], write LDS[k][m] --------
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int ml = sm + 16 * pass;
+      const int64_t m = m0 + ml;
+      uint4 v = zero16();
+      const int k0 = tile_k + su * 8;
+      if (m < m_end && k0 + 8 <= d.K)
+        v = *reinterpret_cast<const uint4*>(dy + m * d.K + k0);
+      const uint16_t* e = reinterpret_cast<const uint16_t*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int row = su * 8 + j;
+        *reinterpret_cast<uint16_t*>(ldsA + swz_off(row, ml * 2)) = e[j];
+      }
+    }
+    // ---- stage x^T: im2col gather, write LDS[c][m] -----------------------
+#pragma unroll
+    for (int pass = 0; pass < 4; ++pass) {
+      const int ml = sm + 16 * pass;
+      const int64_t m = m0 + ml;
+      uint4 v = zero16();
+      uint16_t* e = reinterpret_cast<uint16_t*>(&v);
+      if (m < m_end) {
+        const int q = (int)(m % d.Q);
+        const int p = (int)((m / d.Q) % d.P);
+        const int n = (int)(m / ((int64_t)d.Q * d.P));
+        if (MODE == TILED) {
+          const int hi = p * d.sh + r - d.ph;
+          const int wi = q * d.sw + s - d.pw;
+          const int c0 = tile_c + su * 8;
+          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
+            v = *reinterpret_cast<const uint4*>(
+                x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int kred = tile_c + su * 8 + j;
+            if (kred < kred_total) {
+              const int rr = kred / (d.S * d.C);
+              const int rem = kred % (d.S * d.C);
+              const int ss = rem / d.C;
+              const int cc = rem % d.C;
+              const int hi = p * d.sh + rr - d.ph;
+              const int wi = q * d.sw + ss - d.pw;
+              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                e[j] = *reinterpret_cast<const uint16_t*>(
+                    x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
+            }
+          }
+        }
+      }
+      const uint16_t* ee = reinterpret_cast<const uint16_t*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int row = su * 8 + j;
+        *reinterpret_cast<uint16_t*>(ldsB + swz_off(row, ml * 2)) = ee[j];
+      }
+    }
+    __syncthreads();
+
+    const int frow = lane & 15;
+    const int fcolb = (lane >> 4) * 16;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      typename MfmaFrag<T>::type a[4], b[4];
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr) {
+        uint4 v = lds_read16(ldsA, wr + mr * 16 + frow, kk * 64 + fcolb);
+        a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+      }
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr) {
+        uint4 v = lds_read16(ldsB, wc + nr * 16 + frow, kk * 64 + fcolb);
+        b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+      }
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr)
+          acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: atomicAdd into fp32 dw workspace (KRSC layout) ----------
+  const int en = lane & 15;
+  const int em = (lane >> 4) * 4;
+  const int rsc = d.R * d.S * d.C;
+#pragma unroll
+  for (int mr = 0; mr < 4; ++mr) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int k = tile_k + wr + mr * 16 + em + reg;
+      if (k >= d.K) continue;
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr) {
+        const int cidx = tile_c + wc + nr * 16 + en;
+        if (cidx >= kred_total) continue;
+        const int64_t o = (MODE == FLAT)
+            ? (int64_t)k * rsc + cidx
+            : (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
+        atomicAdd(&dw_ws[o], acc[mr][nr][reg]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, int N,
+                    int H, int W, int C, int K, int R, int S, int P, int Q,
+                    int sh, int sw, int ph, int pw, int dtype, hipStream_t s) {
+  ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
+  const int64_t M = (int64_t)N * P * Q;
+  const bool flat = (C % 8) != 0;
+  const int kred_total = flat ? R * S * C : C;
+  const int ntk = (int)cdiv(K, BM), ntc = (int)cdiv(kred_total, BN);
+  const int rs_total = flat ? 1 : R * S;
+  // split-K over m for parallelism: target >= ~1024 blocks total
+  int64_t chunks = std::max<int64_t>(1, 1024 / ((int64_t)ntk * ntc * rs_total));
+  chunks = std::min<int64_t>(chunks, cdiv(M, BK));
+  int64_t chunk_rows = cdiv(cdiv(M, chunks), BK) * BK;
+  chunks = cdiv(M, chunk_rows);
+  dim3 grid((unsigned)(ntk * ntc), (unsigned)rs_total, (unsigned)chunks);
+#define RUN(T, MODE)                                                           \
+  conv_wgrad_k<T, MODE><<<grid, 256, 0, s>>>((const T*)dy, (const T*)x,        \
+                                             dw_ws, d, M, chunk_rows)
+  if (dtype == 1) { if (flat) RUN(bf16, FLAT); else RUN(bf16, TILED); }
+  else if (dtype == 2) { if (flat) RUN(f16, FLAT); else RUN(f16, TILED); }
+#undef RUN
+}
+
+}  // extern "C"

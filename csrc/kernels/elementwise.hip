@@ -1,0 +1,221 @@
+// Elementwise utilities: ReLU backward mask, finite checks / unscale (K13),
+// dropout (K15, Philox-lite counter RNG), f32 casts, weight transpose (K12).
+// All memory-bound: vectorized 8-wide loads (guide G13), grid-stride.
+#include "common.h"
+#include "api.h"
+
+namespace {
+
+template <typename T>
+__global__ void relu_bwd_k(const T* dy, const T* y, T* out, int64_t n) {
+  int64_t i0 = global_tid() * 8;
+  for (int64_t i = i0; i < n; i += grid_stride() * 8) {
+    if (i + 8 <= n) {
+      Vec8<T> vdy = load8(dy + i), vy = load8(y + i), r;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        r.v[j] = to_f32(vy.v[j]) > 0.f ? vdy.v[j] : from_f32<T>(0.f);
+      store8(out + i, r);
+    } else {
+      for (int64_t k = i; k < n; ++k)
+        out[k] = to_f32(y[k]) > 0.f ? dy[k] : from_f32<T>(0.f);
+    }
+  }
+}
+
+// Bit-level non-finite test (exponent all-ones) — immune to -ffast-math.
+__device__ __forceinline__ bool is_bad(float v) {
+  uint32_t b = __float_as_uint(v);
+  return ((b >> 23) & 0xFFu) == 0xFFu;
+}
+__device__ __forceinline__ bool is_bad(bf16 v) {
+  uint16_t b = *reinterpret_cast<uint16_t*>(&v);
+  return ((b >> 7) & 0xFFu) == 0xFFu;
+}
+__device__ __forceinline__ bool is_bad(f16 v) {
+  uint16_t b = *reinterpret_cast<uint16_t*>(&v);
+  return ((b >> 10) & 0x1Fu) == 0x1Fu;
+}
+
+template <typename T>
+__global__ void check_finite_k(const T* g, int64_t n, int* found_inf) {
+  int bad = 0;
+  int64_t i0 = global_tid() * 8;
+  for (int64_t i = i0; i < n; i += grid_stride() * 8) {
+    if (i + 8 <= n) {
+      Vec8<T> v = load8(g + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) bad |= is_bad(v.v[j]);
+    } else {
+      for (int64_t k = i; k < n; ++k) bad |= is_bad(g[k]);
+    }
+  }
+  if (bad) atomicOr(found_inf, 1);
+}
+
+template <typename T>
+__global__ void unscale_check_k(T* g, int64_t n, float inv_scale,
+                                int* found_inf) {
+  int bad = 0;
+  int64_t i0 = global_tid() * 8;
+  for (int64_t i = i0; i < n; i += grid_stride() * 8) {
+    if (i + 8 <= n) {
+      Vec8<T> v = load8(g + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        bad |= is_bad(v.v[j]);
+        v.v[j] = from_f32<T>(to_f32(v.v[j]) * inv_scale);
+      }
+      store8(g + i, v);
+    } else {
+      for (int64_t k = i; k < n; ++k) {
+        bad |= is_bad(g[k]);
+        g[k] = from_f32<T>(to_f32(g[k]) * inv_scale);
+      }
+    }
+  }
+  if (bad) atomicOr(found_inf, 1);
+}
+
+// Philox-lite 32-bit counter RNG: 4 rounds of the Philox-2x32 round
+// function — statistically fine for dropout masks (not crypto).
+__device__ __forceinline__ uint32_t philox2x32(uint64_t seed, uint64_t ctr) {
+  uint32_t k = (uint32_t)(seed ^ (seed >> 32));
+  uint32_t c0 = (uint32_t)ctr, c1 = (uint32_t)(ctr >> 32);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    uint64_t prod = (uint64_t)0xD256D193u * c0;
+    uint32_t hi = (uint32_t)(prod >> 32), lo = (uint32_t)prod;
+    c0 = hi ^ c1 ^ k;
+    c1 = lo;
+    k += 0x9E3779B9u;
+  }
+  return c0;
+}
+
+template <typename T>
+__global__ void dropout_fwd_k(const T* x, T* y, uint8_t* mask, int64_t n,
+                              float p, uint64_t seed) {
+  float keep_scale = 1.f / (1.f - p);
+  uint32_t thresh = (uint32_t)(p * 4294967296.0f);
+  for (int64_t i = global_tid(); i < n; i += grid_stride()) {
+    uint32_t r = philox2x32(seed, (uint64_t)i);
+    uint8_t keep = r >= thresh;
+    mask[i] = keep;
+    y[i] = keep ? from_f32<T>(to_f32(x[i]) * keep_scale) : from_f32<T>(0.f);
+  }
+}
+
+template <typename T>
+__global__ void dropout_bwd_k(const T* dy, T* dx, const uint8_t* mask,
+                              int64_t n, float p) {
+  float keep_scale = 1.f / (1.f - p);
+  for (int64_t i = global_tid(); i < n; i += grid_stride()) {
+    dx[i] = mask[i] ? from_f32<T>(to_f32(dy[i]) * keep_scale)
+                    : from_f32<T>(0.f);
+  }
+}
+
+template <typename T>
+__global__ void cast_f32_k(const float* src, T* dst, int64_t n) {
+  for (int64_t i = global_tid(); i < n; i += grid_stride())
+    dst[i] = from_f32<T>(src[i]);
+}
+
+// [K, RS, C] -> [C, RS, K]: tiled LDS transpose of the (K, C) planes for
+// each rs.  Grid: (tilesK x tilesC, RS).  32x32 tiles, padded LDS row.
+template <typename T>
+__global__ void transpose_kc_k(const T* w, T* w_t, int K, int RS, int C) {
+  __shared__ T tile[32][33];
+  int rs = blockIdx.y;
+  int ntc = (C + 31) / 32;
+  int tk = (blockIdx.x / ntc) * 32, tc = (blockIdx.x % ntc) * 32;
+  int lx = threadIdx.x % 32, ly = threadIdx.x / 32;  // 32x8 threads
+#pragma unroll
+  for (int yy = 0; yy < 32; yy += 8) {
+    int k = tk + ly + yy, c = tc + lx;
+    tile[ly + yy][lx] = (k < K && c < C)
+        ? w[((int64_t)k * RS + rs) * C + c] : from_f32<T>(0.f);
+  }
+  __syncthreads();
+#pragma unroll
+  for (int yy = 0; yy < 32; yy += 8) {
+    int c = tc + ly + yy, k = tk + lx;
+    if (c < C && k < K)
+      w_t[((int64_t)c * RS + rs) * K + k] = tile[lx][ly + yy];
+  }
+}
+
+inline int ew_grid(int64_t n, int per_thread = 8, int block = 256) {
+  int64_t blocks = cdiv(n, (int64_t)block * per_thread);
+  return (int)std::min<int64_t>(blocks, 2048);
+}
+
+}  // namespace
+
+#define DISPATCH(dtype, F, ...)                                \
+  switch (dtype) {                                             \
+    case 0: F<float>(__VA_ARGS__); break;                      \
+    case 1: F<bf16>(__VA_ARGS__); break;                       \
+    case 2: F<f16>(__VA_ARGS__); break;                        \
+  }
+
+extern "C" {
+
+void ddl_relu_bwd(const void* dy, const void* y, void* out, int64_t n,
+                  int dtype, hipStream_t s) {
+  int g = ew_grid(n);
+#define RUN(T) relu_bwd_k<T><<<g, 256, 0, s>>>((const T*)dy, (const T*)y, (T*)out, n)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_check_finite(const void* g_, int64_t n, int dtype, int* found_inf,
+                      hipStream_t s) {
+  int g = ew_grid(n);
+#define RUN(T) check_finite_k<T><<<g, 256, 0, s>>>((const T*)g_, n, found_inf)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_unscale_check(void* g_, int64_t n, int dtype, float inv_scale,
+                       int* found_inf, hipStream_t s) {
+  int g = ew_grid(n);
+#define RUN(T) unscale_check_k<T><<<g, 256, 0, s>>>((T*)g_, n, inv_scale, found_inf)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_dropout_fwd(const void* x, void* y, uint8_t* mask, int64_t n,
+                     int dtype, float p, uint64_t seed, hipStream_t s) {
+  int g = ew_grid(n, 1);
+#define RUN(T) dropout_fwd_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)y, mask, n, p, seed)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_dropout_bwd(const void* dy, void* dx, const uint8_t* mask, int64_t n,
+                     int dtype, float p, hipStream_t s) {
+  int g = ew_grid(n, 1);
+#define RUN(T) dropout_bwd_k<T><<<g, 256, 0, s>>>((const T*)dy, (T*)dx, mask, n, p)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_cast_f32_to(const float* src, void* dst, int64_t n, int dtype,
+                     hipStream_t s) {
+  int g = ew_grid(n, 4);
+#define RUN(T) cast_f32_k<T><<<g, 256, 0, s>>>(src, (T*)dst, n)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_transpose_krsc_crsk(const void* w, void* w_t, int K, int RS, int C,
+                             int dtype, hipStream_t s) {
+  dim3 grid(cdiv(K, 32) * cdiv(C, 32), RS);
+#define RUN(T) transpose_kc_k<T><<<grid, 256, 0, s>>>((const T*)w, (T*)w_t, K, RS, C)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+}  // extern "C"

@@ -227,10 +227,11 @@ class _SoftmaxXentFn(torch.autograd.Function):
     def backward(ctx, dloss):
         ext = load_kernels_ext()
         probs, target = ctx.saved_tensors
+        code = {torch.float32: 0, torch.bfloat16: 1, torch.float16: 2}[ctx.logits_dtype]
         dlogits = ext.softmax_xent_bwd(probs, target,
                                        float(ctx.label_smoothing),
-                                       dloss / probs.shape[0])
-        return dlogits.to(ctx.logits_dtype), None, None
+                                       dloss.to(torch.float32), code)
+        return dlogits, None, None
 
 
 def softmax_cross_entropy(logits, target, label_smoothing: float = 0.0):

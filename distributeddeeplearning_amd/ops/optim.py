@@ -123,7 +123,9 @@ class FusedSGD(torch.optim.Optimizer):
         dt = group["params"][0].dtype
         if group["lars"]:
             self._norms.zero_()
-            ext.lars_norms(self._table, self._norms, _dtype_code(dt))
+            ext.lars_norms(self._table, self._norms, _dtype_code(dt),
+                           float(group["weight_decay"]),
+                           float(1.0 / self.grad_scale))
         ext.fused_sgd(
             self._table, self._norms if group["lars"] else None,
             _dtype_code(dt),

@@ -1,0 +1,109 @@
+"""Tier-2 single-GPU integration: models train end-to-end on the HIP path,
+loss decreases, checkpoint round-trips through the NCHW/fp32 boundary."""
+import dataclasses
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _trainer(model="resnet18", bs=16, image=64, classes=100, dtype="bf16",
+             **kw):
+    from distributeddeeplearning_amd.config import TrainConfig
+    from distributeddeeplearning_amd.engine.trainer import Trainer
+    cfg = TrainConfig(model=model, image_size=image, num_classes=classes,
+                      batch_size=bs, dtype=dtype, warmup_epochs=0,
+                      comm="single", base_lr=0.05, **kw)
+    return Trainer(cfg, steps_per_epoch=10)
+
+
+def test_resnet18_loss_decreases():
+    tr = _trainer()
+    it = tr.data_iter()
+    losses = []
+    for _ in range(12):
+        x, y = next(it)
+        losses.append(float(tr.train_step(x, y).detach().float()))
+    torch.cuda.synchronize()
+    assert all(l == l for l in losses), "NaN loss"
+    # synthetic data (4 cycling batches) is memorizable: expect clear drop
+    assert min(losses[-4:]) < losses[0], f"no learning: {losses}"
+
+
+def test_resnet50_smoke():
+    tr = _trainer(model="resnet50", bs=8, image=224, classes=1000)
+    it = tr.data_iter()
+    for _ in range(2):
+        x, y = next(it)
+        loss = tr.train_step(x, y)
+    torch.cuda.synchronize()
+    lv = float(loss.detach().float())
+    assert lv == lv and lv < 20
+
+
+def test_vgg16_smoke():
+    tr = _trainer(model="vgg16", bs=4, image=224, classes=1000)
+    it = tr.data_iter()
+    x, y = next(it)
+    loss = tr.train_step(x, y)
+    torch.cuda.synchronize()
+    assert float(loss.detach().float()) == float(loss.detach().float())
+
+
+def test_fp16_loss_scaling_step():
+    tr = _trainer(dtype="fp16", bs=8)
+    it = tr.data_iter()
+    for _ in range(3):
+        x, y = next(it)
+        loss = tr.train_step(x, y)
+    torch.cuda.synchronize()
+    assert tr.scaler.scale > 1.0
+    assert float(loss.detach().float()) == float(loss.detach().float())
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    from distributeddeeplearning_amd.engine import checkpoint as ckpt
+    tr = _trainer(bs=8)
+    it = tr.data_iter()
+    for _ in range(2):
+        x, y = next(it)
+        tr.train_step(x, y)
+    path = str(tmp_path / "ck.pt")
+    ckpt.save(path, tr.model, tr.opt, epoch=1, step=2)
+    payload = torch.load(path, map_location="cpu", weights_only=False)
+    sd = payload["model"]
+    # external layout contract: fp32, NCHW-contiguous, torchvision names
+    w = sd["conv1.weight"]
+    assert w.dtype == torch.float32
+    assert w.is_contiguous()
+    assert "layer1.0.bn1.running_mean" in sd
+
+    tr2 = _trainer(bs=8)
+    ckpt.load(path, tr2.model, tr2.opt)
+    for (k1, v1), (k2, v2) in zip(tr.model.state_dict().items(),
+                                  tr2.model.state_dict().items()):
+        assert k1 == k2
+        if torch.is_tensor(v1) and v1.is_floating_point():
+            assert torch.allclose(v1.float(), v2.float(), atol=1e-2), k1
+
+
+def test_eval_mode_bn_infer():
+    tr = _trainer(bs=8)
+    it = tr.data_iter()
+    for _ in range(3):
+        x, y = next(it)
+        tr.train_step(x, y)
+    tr.model.eval()
+    with torch.no_grad():
+        out = tr.model(x)
+    torch.cuda.synchronize()
+    assert torch.isfinite(out.float()).all()
+
+
+def test_native_extension_is_loaded():
+    """Guard: GPU processes must run our in-tree .so, not eager torch."""
+    import distributeddeeplearning_amd._C_kernels as K
+    assert "distributeddeeplearning_amd" in K.__file__
+    from distributeddeeplearning_amd.ops._ext import load_kernels_ext
+    assert load_kernels_ext() is K
