@@ -120,13 +120,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
               "running stats must stay fp32");
   TORCH_CHECK(weight.scalar_type() == x.scalar_type());
   auto f32 = x.options().dtype(at::kFloat);
-  auto sums = at::zeros({2 * C}, f32);
+  auto sums = at::empty({2 * C}, f32);
+  int chunks = ddl_bn_stats_chunks(M, C);
+  auto partials = at::empty({(int64_t)chunks * 2 * C}, f32);
   auto mean = at::empty({C}, f32);
   auto rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
   auto s = cur_stream();
   int dt = dt_code(x);
-  ddl_bn_stats(x.data_ptr(), M, C, dt, sums.data_ptr<float>(), s);
+  ddl_bn_stats(x.data_ptr(), M, C, dt, sums.data_ptr<float>(),
+               partials.data_ptr<float>(), chunks, s);
   ddl_bn_finalize(sums.data_ptr<float>(), C, M, momentum, eps,
                   rmean.data_ptr<float>(), rvar.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(), 1, s);
@@ -161,10 +164,12 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
   int64_t M = x.numel() / C;
   auto s = cur_stream();
   int dt = dt_code(x);
-  auto sums = at::zeros({2 * C}, x.options().dtype(at::kFloat));
+  auto sums = at::empty({2 * C}, x.options().dtype(at::kFloat));
+  int chunks = ddl_bn_stats_chunks(M, C);
+  auto partials = at::empty({(int64_t)chunks * 2 * C}, x.options().dtype(at::kFloat));
   ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu,
                    mean.data_ptr<float>(), rstd.data_ptr<float>(), M, C, dt,
-                   sums.data_ptr<float>(), s);
+                   sums.data_ptr<float>(), partials.data_ptr<float>(), chunks, s);
   auto dx = at::empty_like(x);
   at::Tensor dres;
   void* dres_p = nullptr;

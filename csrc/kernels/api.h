@@ -35,9 +35,11 @@ void ddl_lars_norms(const int64_t* table, int nrows, float* norms,
                     hipStream_t s);
 
 // ---- fused BatchNorm + residual + ReLU (K4/K5/K14) ------------------------
-// sums: f32 [2*C] workspace (pre-zeroed): 0..C-1 = sum x, C..2C-1 = sum x^2.
+// sums: f32 [2*C]; partials: f32 [chunks*2*C] scratch where chunks =
+// ddl_bn_stats_chunks(M, C).  Two-stage deterministic reduction.
+int ddl_bn_stats_chunks(int64_t M, int C);
 void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
-                  hipStream_t s);
+                  float* partials, int max_chunks, hipStream_t s);
 // finalize: mean/rstd from sums; optionally update running stats in-place.
 void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
                      float eps, float* running_mean, float* running_var,
@@ -57,7 +59,8 @@ void ddl_bn_infer(const void* x, void* y, const void* residual,
 // where dy_m = dy * (y>0) if relu else dy.  Pre-zeroed.
 void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
                       const float* mean, const float* rstd, int64_t M, int C,
-                      int dtype, float* sums, hipStream_t s);
+                      int dtype, float* sums, float* partials, int max_chunks,
+                      hipStream_t s);
 // dx (+ optional dres copy of masked dy), dweight/dbias cast from sums.
 void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y,
                       void* dx, void* dres, const float* mean,

@@ -14,10 +14,13 @@ namespace {
 constexpr int ROWS = 32;     // rows per block pass
 constexpr int CG = 8;        // channel strip = CG vec8-groups = 64 channels
 
-// ---- stats: sums[c] += sum_m x, sums[C+c] += sum_m x^2 --------------------
+// ---- stats: partials[chunk][{0,1}][C] = per-chunk {sum, sumsq} ------------
+// Two-stage (no atomics): 2048 chunks all adding to the same channel would
+// serialize ~2000-deep per address at L2; per-chunk partials + a finalize
+// reduction run at streaming bandwidth AND make BN stats deterministic.
 template <typename T>
 __global__ void bn_stats_k(const T* __restrict__ x, int64_t M, int C,
-                           float* __restrict__ sums) {
+                           float* __restrict__ partials) {
   // block: 256 threads = CG vec-groups x ROWS rows; grid.x = C/64 strips,
   // grid.y = row chunks.
   __shared__ float red[2][ROWS][CG * 8];
@@ -54,12 +57,28 @@ __global__ void bn_stats_k(const T* __restrict__ x, int64_t M, int C,
     __syncthreads();
   }
   if (trow == 0) {
+    float* dst = partials + ((int64_t)blockIdx.y * 2) * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sums[c0 + j], red[0][0][vg * 8 + j]);
-      atomicAdd(&sums[C + c0 + j], red[1][0][vg * 8 + j]);
+      dst[c0 + j] = red[0][0][vg * 8 + j];
+      dst[C + c0 + j] = red[1][0][vg * 8 + j];
     }
   }
+}
+
+// reduce partials[chunks][2][C] into sums[2][C] (one pass, C threads wide)
+__global__ void bn_reduce_partials_k(const float* __restrict__ partials,
+                                     float* __restrict__ sums, int C,
+                                     int chunks) {
+  const int c = threadIdx.x + blockIdx.x * blockDim.x;
+  if (c >= C) return;
+  float s0 = 0.f, s1 = 0.f;
+  for (int i = 0; i < chunks; ++i) {
+    s0 += partials[((int64_t)i * 2) * C + c];
+    s1 += partials[((int64_t)i * 2) * C + C + c];
+  }
+  sums[c] = s0;
+  sums[C + c] = s1;
 }
 
 __global__ void bn_finalize_k(const float* __restrict__ sums, int C,
@@ -190,10 +209,11 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
     __syncthreads();
   }
   if (trow == 0) {
+    float* dst = sums + ((int64_t)blockIdx.y * 2) * C;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sums[c0 + j], red[0][0][vg * 8 + j]);
-      atomicAdd(&sums[C + c0 + j], red[1][0][vg * 8 + j]);
+      dst[c0 + j] = red[0][0][vg * 8 + j];
+      dst[C + c0 + j] = red[1][0][vg * 8 + j];
     }
   }
 }
@@ -260,12 +280,16 @@ inline int apply_grid(int64_t M, int C) {
 
 extern "C" {
 
+int ddl_bn_stats_chunks(int64_t M, int C) { return (int)stats_grid(M, C).y; }
+
 void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
-                  hipStream_t s) {
+                  float* partials, int max_chunks, hipStream_t s) {
   dim3 g = stats_grid(M, C);
-#define RUN(T) bn_stats_k<T><<<g, 256, 0, s>>>((const T*)x, M, C, sums)
+  if ((int)g.y > max_chunks) g.y = max_chunks;
+#define RUN(T) bn_stats_k<T><<<g, 256, 0, s>>>((const T*)x, M, C, partials)
   switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
+  bn_reduce_partials_k<<<cdiv(C, 256), 256, 0, s>>>(partials, sums, C, g.y);
 }
 
 void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
@@ -318,16 +342,19 @@ void ddl_bn_infer(const void* x, void* y, const void* residual,
 
 void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
                       const float* mean, const float* rstd, int64_t M, int C,
-                      int dtype, float* sums, hipStream_t s) {
+                      int dtype, float* sums, float* partials, int max_chunks,
+                      hipStream_t s) {
   dim3 g = stats_grid(M, C);
+  if ((int)g.y > max_chunks) g.y = max_chunks;
 #define RUN(T, RELU)                                                           \
   bn_bwd_stats_k<T, RELU><<<g, 256, 0, s>>>(                                   \
-      (const T*)dy, (const T*)x, (const T*)y, mean, rstd, M, C, sums)
+      (const T*)dy, (const T*)x, (const T*)y, mean, rstd, M, C, partials)
 #define SEL(T)                                                                 \
   if (relu) RUN(T, true); else RUN(T, false)
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN
+  bn_reduce_partials_k<<<cdiv(C, 256), 256, 0, s>>>(partials, sums, C, g.y);
 }
 
 void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y, void* dx,

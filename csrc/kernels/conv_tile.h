@@ -43,8 +43,13 @@ constexpr int BN = 128;   // output-col tile
 constexpr int BK = 64;    // reduce tile
 constexpr int TILE_BYTES = BM * BK * 2;  // 16 KiB per operand tile
 
+// XOR swizzle g(row) = ((row&15) ^ ((row>>3)&7)) << 4.  The (row&15) term
+// makes the b128 fragment reads (16-lane groups, rows m0..m0+15, fixed col)
+// conflict-free; folding in (row>>3)&7 additionally spreads the wgrad
+// staging's b64 transposed writes (rows stride 8 across lanes) to <=2-way.
+// Bits >=4 only, so 16B reads and 8B writes stay aligned; bijective.
 __device__ __forceinline__ int swz_off(int row, int colbyte) {
-  return ((row << 7) + colbyte) ^ ((row & 15) << 4);
+  return ((row << 7) + colbyte) ^ ((((row & 15) ^ ((row >> 3) & 7))) << 4);
 }
 
 __device__ __forceinline__ void lds_write16(char* lds, int row, int colbyte,

@@ -32,10 +32,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   const int wid = threadIdx.x >> 6;
   const int wr = (wid >> 1) * 64;
   const int wc = (wid & 1) * 64;
-  // staging assignment: unit u (0..15) covers 16B of a row of 128 2B els,
-  // m_local = t/16 (0..15) + 16*pass (4 passes -> 64 m)
-  const int su = threadIdx.x & 15;
-  const int sm = threadIdx.x >> 4;
+  // staging assignment (in-register 8x4 transpose, b64 LDS writes): each
+  // thread owns an 8-wide k/c group x 4 consecutive m: 4 x 16B loads ->
+  // repack -> 8 x 8B transposed writes.  16 k-groups x 16 m-groups = 256.
+  const int su = threadIdx.x & 15;        // k/c group (0..15) -> 8 rows
+  const int sm4 = (threadIdx.x >> 4) * 4; // first of 4 owned m columns
 
   f32x4_t acc[4][4] = {};
 
@@ -43,63 +44,75 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   const int64_t m_end = min(M, m_begin + chunk_rows);
 
   for (int64_t m0 = m_begin; m0 < m_end; m0 += BK) {
-    // ---- stage dy^T: load dy[m][tile_k + u*8 ..], write LDS[k][m] --------
+    // ---- stage dy^T: load 4 m-rows of 8 k, write LDS[k][m] b64 -----------
+    {
+      uint4 v[4];
+      const int k0g = tile_k + su * 8;
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
-      const int ml = sm + 16 * pass;
-      const int64_t m = m0 + ml;
-      uint4 v = zero16();
-      const int k0 = tile_k + su * 8;
-      if (m < m_end && k0 + 8 <= d.K)
-        v = *reinterpret_cast<const uint4*>(dy + m * d.K + k0);
-      const uint16_t* e = reinterpret_cast<const uint16_t*>(&v);
+      for (int mm = 0; mm < 4; ++mm) {
+        const int64_t m = m0 + sm4 + mm;
+        v[mm] = zero16();
+        if (m < m_end && k0g + 8 <= d.K)
+          v[mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
+      }
+      const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
+      const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
+      const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
+      const uint16_t* e3 = reinterpret_cast<const uint16_t*>(&v[3]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int row = su * 8 + j;
-        *reinterpret_cast<uint16_t*>(ldsA + swz_off(row, ml * 2)) = e[j];
+        uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
+                      ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
+        *reinterpret_cast<uint64_t*>(ldsA + swz_off(su * 8 + j, sm4 * 2)) = pk;
       }
     }
-    // ---- stage x^T: im2col gather, write LDS[c][m] -----------------------
+    // ---- stage x^T: im2col gather, same transposed write -----------------
+    {
+      uint4 v[4];
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
-      const int ml = sm + 16 * pass;
-      const int64_t m = m0 + ml;
-      uint4 v = zero16();
-      uint16_t* e = reinterpret_cast<uint16_t*>(&v);
-      if (m < m_end) {
-        const int q = (int)(m % d.Q);
-        const int p = (int)((m / d.Q) % d.P);
-        const int n = (int)(m / ((int64_t)d.Q * d.P));
-        if (MODE == TILED) {
-          const int hi = p * d.sh + r - d.ph;
-          const int wi = q * d.sw + s - d.pw;
-          const int c0 = tile_c + su * 8;
-          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
-            v = *reinterpret_cast<const uint4*>(
-                x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
-        } else {
+      for (int mm = 0; mm < 4; ++mm) {
+        const int64_t m = m0 + sm4 + mm;
+        v[mm] = zero16();
+        uint16_t* e = reinterpret_cast<uint16_t*>(&v[mm]);
+        if (m < m_end) {
+          const int q = (int)(m % d.Q);
+          const int p = (int)((m / d.Q) % d.P);
+          const int n = (int)(m / ((int64_t)d.Q * d.P));
+          if (MODE == TILED) {
+            const int hi = p * d.sh + r - d.ph;
+            const int wi = q * d.sw + s - d.pw;
+            const int c0 = tile_c + su * 8;
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
+              v[mm] = *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
+          } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int kred = tile_c + su * 8 + j;
-            if (kred < kred_total) {
-              const int rr = kred / (d.S * d.C);
-              const int rem = kred % (d.S * d.C);
-              const int ss = rem / d.C;
-              const int cc = rem % d.C;
-              const int hi = p * d.sh + rr - d.ph;
-              const int wi = q * d.sw + ss - d.pw;
-              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                e[j] = *reinterpret_cast<const uint16_t*>(
-                    x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
+            for (int j = 0; j < 8; ++j) {
+              const int kred = tile_c + su * 8 + j;
+              if (kred < kred_total) {
+                const int rr = kred / (d.S * d.C);
+                const int rem = kred % (d.S * d.C);
+                const int ss = rem / d.C;
+                const int cc = rem % d.C;
+                const int hi = p * d.sh + rr - d.ph;
+                const int wi = q * d.sw + ss - d.pw;
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                  e[j] = *reinterpret_cast<const uint16_t*>(
+                      x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
+              }
             }
           }
         }
       }
-      const uint16_t* ee = reinterpret_cast<const uint16_t*>(&v);
+      const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
+      const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
+      const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
+      const uint16_t* e3 = reinterpret_cast<const uint16_t*>(&v[3]);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int row = su * 8 + j;
-        *reinterpret_cast<uint16_t*>(ldsB + swz_off(row, ml * 2)) = ee[j];
+        uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
+                      ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
+        *reinterpret_cast<uint64_t*>(ldsB + swz_off(su * 8 + j, sm4 * 2)) = pk;
       }
     }
     __syncthreads();
