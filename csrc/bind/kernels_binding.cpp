@@ -96,13 +96,12 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int K = w_shape[0], R = w_shape[2], S = w_shape[3];
   int P = dy.size(2), Q = dy.size(3);
-  auto ws = at::zeros({(int64_t)K * R * S * C},
+  auto ws = at::empty({ddl_conv_wgrad_ws_elems(N, C, K, R, S, P, Q)},
                       x.options().dtype(at::kFloat));
-  ddl_conv_wgrad(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(), N, H, W,
-                 C, K, R, S, P, Q, sh, sw, ph, pw, dt_code(x), cur_stream());
   auto dw = empty_nhwc(K, C, R, S, x.options());
-  ddl_cast_f32_to(ws.data_ptr<float>(), dw.data_ptr(), (int64_t)K * R * S * C,
-                  dt_code(x), cur_stream());
+  ddl_conv_wgrad(dy.data_ptr(), x.data_ptr(), ws.data_ptr<float>(),
+                 dw.data_ptr(), N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw,
+                 dt_code(x), cur_stream());
   return dw;
 }
 

@@ -100,10 +100,14 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
 void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
                     int W, int C, int K, int R, int S, int P, int Q, int sh,
                     int sw, int ph, int pw, int dtype, hipStream_t s);
-// wgrad accumulates fp32 into dw_ws [K,R,S,C] (pre-zeroed).
-void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, int N,
-                    int H, int W, int C, int K, int R, int S, int P, int Q,
-                    int sh, int sw, int ph, int pw, int dtype, hipStream_t s);
+// wgrad: fp32 per-chunk slabs in dw_ws (ddl_conv_wgrad_ws_elems floats,
+// no pre-zero needed) reduced+cast into dw (KRSC, dtype).  Deterministic.
+int64_t ddl_conv_wgrad_ws_elems(int N, int C, int K, int R, int S, int P,
+                                int Q);
+void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
+                    int N, int H, int W, int C, int K, int R, int S, int P,
+                    int Q, int sh, int sw, int ph, int pw, int dtype,
+                    hipStream_t s);
 // [K, RS, C] -> [C, RS, K] element transpose (K12 utility).
 void ddl_transpose_krsc_crsk(const void* w, void* w_t, int K, int RS, int C,
                              int dtype, hipStream_t s);

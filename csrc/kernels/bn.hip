@@ -66,19 +66,51 @@ __global__ void bn_stats_k(const T* __restrict__ x, int64_t M, int C,
   }
 }
 
-// reduce partials[chunks][2][C] into sums[2][C] (one pass, C threads wide)
+// reduce partials[chunks][2][C] into sums[2][C]: block = 256 threads =
+// 8 channel-vec-groups x 32 chunk-rows (same shape as the stats kernel so
+// the reads stay coalesced and the reduction is parallel over chunks).
 __global__ void bn_reduce_partials_k(const float* __restrict__ partials,
                                      float* __restrict__ sums, int C,
                                      int chunks) {
-  const int c = threadIdx.x + blockIdx.x * blockDim.x;
-  if (c >= C) return;
-  float s0 = 0.f, s1 = 0.f;
-  for (int i = 0; i < chunks; ++i) {
-    s0 += partials[((int64_t)i * 2) * C + c];
-    s1 += partials[((int64_t)i * 2) * C + C + c];
+  __shared__ float red[2][ROWS][CG * 8];
+  const int vg = threadIdx.x % CG;
+  const int trow = threadIdx.x / CG;
+  const int c0 = blockIdx.x * CG * 8 + vg * 8;
+  if (c0 >= C) return;
+  float s0[8] = {0}, s1[8] = {0};
+  for (int i = trow; i < chunks; i += ROWS) {
+    const float* p0 = partials + ((int64_t)i * 2) * C + c0;
+    const float* p1 = p0 + C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      s0[j] += p0[j];
+      s1[j] += p1[j];
+    }
   }
-  sums[c] = s0;
-  sums[C + c] = s1;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][trow][vg * 8 + j] = s0[j];
+    red[1][trow][vg * 8 + j] = s1[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int step = ROWS / 2; step > 0; step >>= 1) {
+    if (trow < step) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        red[0][trow][vg * 8 + j] += red[0][trow + step][vg * 8 + j];
+        red[1][trow][vg * 8 + j] += red[1][trow + step][vg * 8 + j];
+      }
+    }
+    __syncthreads();
+  }
+  if (trow == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      sums[c0 + j] = red[0][0][vg * 8 + j];
+      sums[C + c0 + j] = red[1][0][vg * 8 + j];
+    }
+  }
 }
 
 __global__ void bn_finalize_k(const float* __restrict__ sums, int C,
@@ -289,7 +321,7 @@ void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
 #define RUN(T) bn_stats_k<T><<<g, 256, 0, s>>>((const T*)x, M, C, partials)
   switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
-  bn_reduce_partials_k<<<cdiv(C, 256), 256, 0, s>>>(partials, sums, C, g.y);
+  bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(partials, sums, C, g.y);
 }
 
 void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
@@ -354,7 +386,7 @@ void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN
-  bn_reduce_partials_k<<<cdiv(C, 256), 256, 0, s>>>(partials, sums, C, g.y);
+  bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(partials, sums, C, g.y);
 }
 
 void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y, void* dx,

@@ -5,14 +5,15 @@
 // Dgrad:    dx[m=(n,h,w)][c] = sum_{r,s,k} dy[n,(h+ph-r)/sh,(w+pw-s)/sw,k] *
 //                                          w[k,r,s,c]   (terms where /sh,/sw
 //                                          divide exactly)
-// Both are the same GEMM loop: per (r,s), per 64-wide reduce tile, stage an
-// im2col A-tile (hardware-free zero padding via validity masks) and a
-// B^T-layout weight tile into swizzled LDS, then 2x16 MFMA per wave.
-// Dgrad consumes the CRSK-transposed weight so its B-tile loads are
-// contiguous (built once per backward by ddl_transpose_krsc_crsk).
-//
-// MODE_FLAT handles C not divisible by 8 (conv1: C=3) by flattening the
-// reduce dim to R*S*C with per-element scalar gather.
+// Both are the same GEMM pipeline over the linearized (r,s,c-tile) reduce
+// sequence: double-buffered swizzled LDS tiles with register staging split
+// issue-early/write-late (guide §6 G15 T14: next tile's global loads issue
+// before the MFMA section, the LDS write lands after it, one barrier per
+// tile), 128x128x64 block tile, 4 waves, v_mfma_f32_16x16x32_{bf16,f16}.
+// Padding is handled by per-row validity masks (zero-fill); dgrad consumes
+// the CRSK-transposed weight (ddl_transpose_krsc_crsk) so its B-tile loads
+// are contiguous.  MODE_FLAT flattens the reduce dim to R*S*C with scalar
+// gather for C % 8 != 0 (the 7x7 C=3 stem).
 #include "conv_tile.h"
 #include "api.h"
 
@@ -25,9 +26,9 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
     int Nc, int relu) {
-  __shared__ __align__(16) char smem[2 * TILE_BYTES];
-  char* ldsA = smem;
-  char* ldsB = smem + TILE_BYTES;
+  __shared__ __align__(16) char smem[4 * TILE_BYTES];  // A0 B0 A1 B1
+  char* const lA[2] = {smem, smem + 2 * TILE_BYTES};
+  char* const lB[2] = {smem + TILE_BYTES, smem + 3 * TILE_BYTES};
 
   const int nt_n = (Nc + BN - 1) / BN;
   const unsigned bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -36,16 +37,15 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wr = (wid >> 1) * 64;  // wave row offset in tile
-  const int wc = (wid & 1) * 64;   // wave col offset
-  const int trow = threadIdx.x >> 3;  // staging row base (0..31)
-  const int tcol = threadIdx.x & 7;   // staging 16B unit (0..7)
+  const int wr = (wid >> 1) * 64;
+  const int wc = (wid & 1) * 64;
+  const int trow = threadIdx.x >> 3;
+  const int tcol = threadIdx.x & 7;
 
   f32x4_t acc[4][4] = {};
 
-  // -- per-thread A-row decode (4 staged rows) ------------------------------
-  int64_t a_row_off[4];  // offset of the (n,hi,wi) pixel * C, or -1 invalid
-  int a_n[4], a_i1[4], a_i2[4];  // decoded (n, p|h, q|w) per staged row
+  // per-thread A-row decode (4 staged rows)
+  int a_n[4], a_i1[4], a_i2[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     int64_t m = tile_m + trow + 32 * i;
@@ -64,134 +64,151 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
 
-  const int Cred = (MODE == DGRAD) ? d.K : d.C;      // reduce length per rs
-  const int64_t a_ld = (MODE == DGRAD) ? d.K : d.C;  // A row stride (els)
-  const int b_ld = (MODE == DGRAD) ? d.K : d.C;      // B rs-slice row stride
-
+  const int Cred = (MODE == DGRAD) ? d.K : d.C;
+  const int64_t a_ld = (MODE == DGRAD) ? d.K : d.C;
+  const int b_ld = (MODE == DGRAD) ? d.K : d.C;
   const int rs_total = (MODE == FWD_FLAT) ? 1 : d.R * d.S;
   const int kred_total = (MODE == FWD_FLAT) ? d.R * d.S * d.C : Cred;
+  const int nct = (kred_total + BK - 1) / BK;
+  const int total_tiles = rs_total * nct;
 
-  for (int rs = 0; rs < rs_total; ++rs) {
+  int cur_rs = -1;
+  const T* a_ptr[4];
+  const T* b_ptr[4];
+
+  // ---- per-tile register staging (issue-early half) ----------------------
+  uint4 ra[4], rb[4];
+  auto load_tile = [&](int t) {
+    const int rs = t / nct;
+    const int c0 = (t % nct) * BK;
     const int r = rs / d.S, s = rs % d.S;
-    // Per-rs A row pointers + validity
-    const T* a_ptr[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      a_ptr[i] = nullptr;
-      if (a_n[i] < 0) continue;
-      if (MODE == FWD) {
-        const int hi = a_i1[i] * d.sh + r - d.ph;
-        const int wi = a_i2[i] * d.sw + s - d.pw;
-        if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-          a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
-      } else if (MODE == DGRAD) {
-        const int hn = a_i1[i] + d.ph - r;
-        const int wn = a_i2[i] + d.pw - s;
-        if (hn >= 0 && wn >= 0 && hn % d.sh == 0 && wn % d.sw == 0) {
-          const int ho = hn / d.sh, wo = wn / d.sw;
-          if (ho < d.P && wo < d.Q)
-            a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
-        }
-      } else {  // FWD_FLAT: row pixel base only (per-element taps)
-        a_ptr[i] = Asrc;  // marker: row exists
-      }
-    }
-    // B row (output-col) pointers for this rs
-    // fwd: row = k_out -> w[((k*R + r)*S + s)*C + c]
-    // dgrad: row = c_in -> w_t[((c*R + r)*S + s)*K + k]
-    const T* b_ptr[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int row_n = tile_n + trow + 32 * i;
-      b_ptr[i] = nullptr;
-      if (row_n < Nc) {
-        if (MODE == FWD_FLAT)
-          b_ptr[i] = Bsrc + (int64_t)row_n * kred_total;
-        else
-          b_ptr[i] = Bsrc + ((int64_t)row_n * d.R + r) * d.S * b_ld +
-                     (int64_t)s * b_ld;
-      }
-    }
-
-    for (int c0 = 0; c0 < kred_total; c0 += BK) {
-      // ---- stage A ----
+    if (rs != cur_rs) {
+      cur_rs = rs;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        const int row = trow + 32 * i;
-        const int cel = c0 + tcol * 8;
-        uint4 v = zero16();
-        if (MODE == FWD_FLAT) {
-          if (a_n[i] >= 0) {
-            uint16_t* e = reinterpret_cast<uint16_t*>(&v);
+        a_ptr[i] = nullptr;
+        if (a_n[i] >= 0) {
+          if (MODE == FWD) {
+            const int hi = a_i1[i] * d.sh + r - d.ph;
+            const int wi = a_i2[i] * d.sw + s - d.pw;
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+              a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
+          } else if (MODE == DGRAD) {
+            const int hn = a_i1[i] + d.ph - r;
+            const int wn = a_i2[i] + d.pw - s;
+            if (hn >= 0 && wn >= 0 && hn % d.sh == 0 && wn % d.sw == 0) {
+              const int ho = hn / d.sh, wo = wn / d.sw;
+              if (ho < d.P && wo < d.Q)
+                a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
+            }
+          } else {
+            a_ptr[i] = Asrc;
+          }
+        }
+      }
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int kred = cel + j;
-              if (kred < kred_total) {
-                const int rr = kred / (d.S * d.C);
-                const int rem = kred % (d.S * d.C);
-                const int ss = rem / d.C;
-                const int cc = rem % d.C;
-                const int hi = a_i1[i] * d.sh + rr - d.ph;
-                const int wi = a_i2[i] * d.sw + ss - d.pw;
-                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W) {
-                  const T* p = Asrc +
-                      (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc;
-                  e[j] = *reinterpret_cast<const uint16_t*>(p);
-                }
-              }
+      for (int i = 0; i < 4; ++i) {
+        const int row_n = tile_n + trow + 32 * i;
+        b_ptr[i] = nullptr;
+        if (row_n < Nc) {
+          if (MODE == FWD_FLAT)
+            b_ptr[i] = Bsrc + (int64_t)row_n * kred_total;
+          else
+            b_ptr[i] = Bsrc + ((int64_t)row_n * d.R + r) * d.S * b_ld +
+                       (int64_t)s * b_ld;
+        }
+      }
+    }
+    const int cel = c0 + tcol * 8;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      ra[i] = zero16();
+      if (MODE == FWD_FLAT) {
+        if (a_n[i] >= 0) {
+          uint16_t* e = reinterpret_cast<uint16_t*>(&ra[i]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int kred = cel + j;
+            if (kred < kred_total) {
+              const int rr = kred / (d.S * d.C);
+              const int rem = kred % (d.S * d.C);
+              const int ss = rem / d.C;
+              const int cc = rem % d.C;
+              const int hi = a_i1[i] * d.sh + rr - d.ph;
+              const int wi = a_i2[i] * d.sw + ss - d.pw;
+              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                e[j] = *reinterpret_cast<const uint16_t*>(
+                    Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc);
             }
           }
-        } else {
-          if (a_ptr[i] != nullptr && cel + 8 <= Cred)
-            v = *reinterpret_cast<const uint4*>(a_ptr[i] + cel);
         }
-        lds_write16(ldsA, row, tcol * 16, v);
+      } else if (a_ptr[i] != nullptr && cel + 8 <= Cred) {
+        ra[i] = *reinterpret_cast<const uint4*>(a_ptr[i] + cel);
       }
-      // ---- stage B ----
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row = trow + 32 * i;
-        const int cel = c0 + tcol * 8;
-        uint4 v = zero16();
-        if (b_ptr[i] != nullptr) {
-          if (MODE == FWD_FLAT) {
-            uint16_t* e = reinterpret_cast<uint16_t*>(&v);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              if (cel + j < kred_total)
-                e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
-          } else if (cel + 8 <= Cred) {
-            v = *reinterpret_cast<const uint4*>(b_ptr[i] + cel);
-          }
-        }
-        lds_write16(ldsB, row, tcol * 16, v);
-      }
-      __syncthreads();
-
-      // ---- MFMA ----
-      const int frow = lane & 15;
-      const int fcolb = (lane >> 4) * 16;  // byte offset of this lane's 8 els
-#pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        typename MfmaFrag<T>::type a[4], b[4];
-#pragma unroll
-        for (int mr = 0; mr < 4; ++mr) {
-          uint4 v = lds_read16(ldsA, wr + mr * 16 + frow, kk * 64 + fcolb);
-          a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-        }
-#pragma unroll
-        for (int nr = 0; nr < 4; ++nr) {
-          uint4 v = lds_read16(ldsB, wc + nr * 16 + frow, kk * 64 + fcolb);
-          b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-        }
-#pragma unroll
-        for (int mr = 0; mr < 4; ++mr)
-#pragma unroll
-          for (int nr = 0; nr < 4; ++nr)
-            acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
-      }
-      __syncthreads();
     }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      rb[i] = zero16();
+      if (b_ptr[i] != nullptr) {
+        if (MODE == FWD_FLAT) {
+          uint16_t* e = reinterpret_cast<uint16_t*>(&rb[i]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (cel + j < kred_total)
+              e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
+        } else if (cel + 8 <= Cred) {
+          rb[i] = *reinterpret_cast<const uint4*>(b_ptr[i] + cel);
+        }
+      }
+    }
+  };
+
+  auto write_tile = [&](char* la, char* lb) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      lds_write16(la, trow + 32 * i, tcol * 16, ra[i]);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      lds_write16(lb, trow + 32 * i, tcol * 16, rb[i]);
+  };
+
+  const int frow = lane & 15;
+  const int fcolb = (lane >> 4) * 16;
+  auto mfma_tile = [&](const char* la, const char* lb) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      typename MfmaFrag<T>::type a[4], b[4];
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr) {
+        uint4 v = lds_read16(la, wr + mr * 16 + frow, kk * 64 + fcolb);
+        a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+      }
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr) {
+        uint4 v = lds_read16(lb, wc + nr * 16 + frow, kk * 64 + fcolb);
+        b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+      }
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr)
+          acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+    }
+  };
+
+  // ---- pipelined main loop: one barrier per tile -------------------------
+  load_tile(0);
+  write_tile(lA[0], lB[0]);
+  __syncthreads();
+  for (int t = 0; t < total_tiles; ++t) {
+    const int cur = t & 1;
+    const bool more = (t + 1) < total_tiles;
+    if (more) load_tile(t + 1);                 // global loads in flight...
+    char* la = cur ? lA[1] : lA[0];
+    char* lb = cur ? lB[1] : lB[0];
+    mfma_tile(la, lb);                          // ...hidden under the MFMAs
+    if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+    __syncthreads();
   }
 
   // ---- epilogue: D[m][n], m = (lane>>4)*4 + reg, n = lane&15 --------------

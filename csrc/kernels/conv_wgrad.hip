@@ -1,11 +1,11 @@
 // K3: wgrad implicit GEMM — dw[k,r,s,c] = sum_m dy[m,k] * im2col_x[m,(r,s,c)].
 // Per (r,s) and m-chunk: GEMM tile dw[128k][128c], reduce over the chunk's
 // rows 64 at a time.  Both operands need the reduce dim (m) innermost in the
-// fragment, so dy and x tiles are TRANSPOSE-staGED into LDS ([k][m] / [c][m])
-// with scalar ds_write_b16 (v1; tr_b16 reads are the planned upgrade), then
-// the MFMA loop reads b128 like the forward kernel.  Chunks accumulate into
-// an fp32 workspace with atomicAdd (split-K over m; order nondeterministic
-// in the low bits, as cuDNN's default wgrad is).
+// fragment, so dy and x tiles are TRANSPOSE-STAGED into LDS ([k][m] / [c][m])
+// via an in-register 8x4 repack (4 coalesced 16B loads -> 8 b64 writes,
+// bank-spread by the shared swizzle), then the MFMA loop reads b128 like the
+// forward kernel.  Split-K over m writes per-chunk fp32 slabs (plain dense
+// stores, no atomics) combined by a deterministic reduce+cast kernel.
 #include "conv_tile.h"
 #include "api.h"
 
@@ -16,7 +16,12 @@ enum WMode { TILED = 0, FLAT = 1 };
 template <typename T, int MODE>
 __global__ __launch_bounds__(256) void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
-    float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows) {
+    float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows,
+    int64_t slab_len) {
+  // each m-chunk (blockIdx.z) owns a private fp32 slab: plain dense stores,
+  // no atomics (a chunks-deep atomicAdd chain per dw element serializes at
+  // L2 like the BN-stats case), and the final sum is deterministic.
+  dw_ws += (int64_t)blockIdx.z * slab_len;
   __shared__ __align__(16) char smem[2 * TILE_BYTES];
   char* ldsA = smem;                 // dy^T tile: [128 k][64 m]
   char* ldsB = smem + TILE_BYTES;    // x^T tile:  [128 c|kred][64 m]
@@ -141,7 +146,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
     __syncthreads();
   }
 
-  // ---- epilogue: atomicAdd into fp32 dw workspace (KRSC layout) ----------
+  // ---- epilogue: plain dense stores into this chunk's fp32 slab ----------
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
   const int rsc = d.R * d.S * d.C;
@@ -158,37 +163,87 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
         const int64_t o = (MODE == FLAT)
             ? (int64_t)k * rsc + cidx
             : (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
-        atomicAdd(&dw_ws[o], acc[mr][nr][reg]);
+        dw_ws[o] = acc[mr][nr][reg];
       }
     }
   }
+}
+
+// dw[i] = cast(sum_ch ws[ch][i]) — the deterministic chunk combine.
+template <typename T>
+__global__ void wgrad_reduce_cast_k(const float* __restrict__ ws, T* dw,
+                                    int64_t len, int chunks) {
+  for (int64_t i = global_tid() * 4; i < len; i += grid_stride() * 4) {
+#pragma unroll
+    for (int jj = 0; jj < 4; ++jj) {
+      const int64_t o = i + jj;
+      if (o >= len) break;
+      float acc = 0.f;
+      for (int ch = 0; ch < chunks; ++ch) acc += ws[(int64_t)ch * len + o];
+      dw[o] = from_f32<T>(acc);
+    }
+  }
+}
+
+struct WgradPlan {
+  int64_t chunks, chunk_rows;
+  int ntk, ntc, rs_total, kred_total;
+};
+
+WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
+  WgradPlan pl;
+  const int64_t M = (int64_t)N * P * Q;
+  const bool flat = (C % 8) != 0;
+  pl.kred_total = flat ? R * S * C : C;
+  pl.ntk = (int)cdiv(K, BM);
+  pl.ntc = (int)cdiv(pl.kred_total, BN);
+  pl.rs_total = flat ? 1 : R * S;
+  int64_t chunks = std::max<int64_t>(
+      1, 1024 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
+  chunks = std::min<int64_t>(chunks, cdiv(M, BK));
+  // cap slab memory at ~256 MB
+  const int64_t slab = (int64_t)K * R * S * C * 4;
+  if (slab > 0) chunks = std::min<int64_t>(chunks, std::max<int64_t>(1, (256LL << 20) / slab));
+  pl.chunk_rows = cdiv(cdiv(M, chunks), BK) * BK;
+  pl.chunks = cdiv(M, pl.chunk_rows);
+  return pl;
 }
 
 }  // namespace
 
 extern "C" {
 
-void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, int N,
-                    int H, int W, int C, int K, int R, int S, int P, int Q,
-                    int sh, int sw, int ph, int pw, int dtype, hipStream_t s) {
+int64_t ddl_conv_wgrad_ws_elems(int N, int C, int K, int R, int S, int P,
+                                int Q) {
+  WgradPlan pl = wgrad_plan(N, C, K, R, S, P, Q);
+  return pl.chunks * (int64_t)K * R * S * C;
+}
+
+void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
+                    int N, int H, int W, int C, int K, int R, int S, int P,
+                    int Q, int sh, int sw, int ph, int pw, int dtype,
+                    hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int64_t M = (int64_t)N * P * Q;
   const bool flat = (C % 8) != 0;
-  const int kred_total = flat ? R * S * C : C;
-  const int ntk = (int)cdiv(K, BM), ntc = (int)cdiv(kred_total, BN);
-  const int rs_total = flat ? 1 : R * S;
-  // split-K over m for parallelism: target >= ~1024 blocks total
-  int64_t chunks = std::max<int64_t>(1, 1024 / ((int64_t)ntk * ntc * rs_total));
-  chunks = std::min<int64_t>(chunks, cdiv(M, BK));
-  int64_t chunk_rows = cdiv(cdiv(M, chunks), BK) * BK;
-  chunks = cdiv(M, chunk_rows);
-  dim3 grid((unsigned)(ntk * ntc), (unsigned)rs_total, (unsigned)chunks);
+  WgradPlan pl = wgrad_plan(N, C, K, R, S, P, Q);
+  const int64_t slab_len = (int64_t)K * R * S * C;
+  dim3 grid((unsigned)(pl.ntk * pl.ntc), (unsigned)pl.rs_total,
+            (unsigned)pl.chunks);
 #define RUN(T, MODE)                                                           \
   conv_wgrad_k<T, MODE><<<grid, 256, 0, s>>>((const T*)dy, (const T*)x,        \
-                                             dw_ws, d, M, chunk_rows)
+                                             dw_ws, d, M, pl.chunk_rows,       \
+                                             slab_len)
   if (dtype == 1) { if (flat) RUN(bf16, FLAT); else RUN(bf16, TILED); }
   else if (dtype == 2) { if (flat) RUN(f16, FLAT); else RUN(f16, TILED); }
 #undef RUN
+  int g = (int)std::min<int64_t>(cdiv(slab_len, 1024), 2048);
+#define RED(T)                                                                 \
+  wgrad_reduce_cast_k<T><<<g, 256, 0, s>>>(dw_ws, (T*)dw, slab_len,            \
+                                           (int)pl.chunks)
+  if (dtype == 1) RED(bf16);
+  else if (dtype == 2) RED(f16);
+#undef RED
 }
 
 }  // extern "C"
