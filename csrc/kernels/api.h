@@ -108,6 +108,15 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
                     int N, int H, int W, int C, int K, int R, int S, int P,
                     int Q, int sh, int sw, int ph, int pw, int dtype,
                     hipStream_t s);
+// im2col (small-C stem): x -> [N*P*Q, ldout] with (r,s,c) flattened cols
+// zero-padded to ldout (multiple of 8); plus K x kred <-> K x ldout pads.
+void ddl_im2col(const void* x, void* out, int N, int H, int W, int C, int R,
+                int S, int P, int Q, int sh, int sw, int ph, int pw,
+                int ldout, int dtype, hipStream_t s);
+void ddl_pad_cols(const void* w, void* out, int K, int kred, int ldout,
+                  int dtype, hipStream_t s);
+void ddl_narrow_cols(const void* w, void* out, int K, int kred, int ldout,
+                     int dtype, hipStream_t s);
 // [K, RS, C] -> [C, RS, K] element transpose (K12 utility).
 void ddl_transpose_krsc_crsk(const void* w, void* w_t, int K, int RS, int C,
                              int dtype, hipStream_t s);

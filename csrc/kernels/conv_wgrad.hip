@@ -170,18 +170,37 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
 }
 
 // dw[i] = cast(sum_ch ws[ch][i]) — the deterministic chunk combine.
+// Block = 64 float4-element-groups x 4 chunk-quarters, so the chunk loop is
+// 4-way parallel and the grid covers the dw plane densely even when it is
+// small (a serial per-thread chunk loop left 95% of the chip idle).
 template <typename T>
 __global__ void wgrad_reduce_cast_k(const float* __restrict__ ws, T* dw,
                                     int64_t len, int chunks) {
-  for (int64_t i = global_tid() * 4; i < len; i += grid_stride() * 4) {
-#pragma unroll
-    for (int jj = 0; jj < 4; ++jj) {
-      const int64_t o = i + jj;
-      if (o >= len) break;
-      float acc = 0.f;
-      for (int ch = 0; ch < chunks; ++ch) acc += ws[(int64_t)ch * len + o];
-      dw[o] = from_f32<T>(acc);
+  __shared__ float red[4][64][4];
+  const int eg = threadIdx.x & 63;   // element group (4 floats)
+  const int cq = threadIdx.x >> 6;   // chunk quarter
+  const int64_t e0 = ((int64_t)blockIdx.x * 64 + eg) * 4;
+  float acc[4] = {0, 0, 0, 0};
+  if (e0 < len) {
+    for (int ch = cq; ch < chunks; ch += 4) {
+      const float* p = ws + (int64_t)ch * len + e0;
+      if (e0 + 4 <= len) {
+        float4 v = *reinterpret_cast<const float4*>(p);
+        acc[0] += v.x; acc[1] += v.y; acc[2] += v.z; acc[3] += v.w;
+      } else {
+        for (int j = 0; j < (int)(len - e0); ++j) acc[j] += p[j];
+      }
     }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) red[cq][eg][j] = acc[j];
+  __syncthreads();
+  if (cq == 0 && e0 < len) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      acc[j] = red[0][eg][j] + red[1][eg][j] + red[2][eg][j] + red[3][eg][j];
+    const int nt = (int)min((int64_t)4, len - e0);
+    for (int j = 0; j < nt; ++j) dw[e0 + j] = from_f32<T>(acc[j]);
   }
 }
 
@@ -199,8 +218,9 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   pl.ntc = (int)cdiv(pl.kred_total, BN);
   pl.rs_total = flat ? 1 : R * S;
   int64_t chunks = std::max<int64_t>(
-      1, 1024 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
+      1, 768 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
   chunks = std::min<int64_t>(chunks, cdiv(M, BK));
+  chunks = std::min<int64_t>(chunks, 128);  // bound slab traffic
   // cap slab memory at ~256 MB
   const int64_t slab = (int64_t)K * R * S * C * 4;
   if (slab > 0) chunks = std::min<int64_t>(chunks, std::max<int64_t>(1, (256LL << 20) / slab));
@@ -237,7 +257,7 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   if (dtype == 1) { if (flat) RUN(bf16, FLAT); else RUN(bf16, TILED); }
   else if (dtype == 2) { if (flat) RUN(f16, FLAT); else RUN(f16, TILED); }
 #undef RUN
-  int g = (int)std::min<int64_t>(cdiv(slab_len, 1024), 2048);
+  int g = (int)cdiv(slab_len, 256);
 #define RED(T)                                                                 \
   wgrad_reduce_cast_k<T><<<g, 256, 0, s>>>(dw_ws, (T*)dw, slab_len,            \
                                            (int)pl.chunks)

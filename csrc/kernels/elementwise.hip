@@ -146,6 +146,67 @@ __global__ void transpose_kc_k(const T* w, T* w_t, int K, int RS, int C) {
   }
 }
 
+// im2col for small-C stems (conv1: 7x7 s2 C=3): rows m=(n,p,q), cols
+// kred=(r,s,c) zero-padded to `ldout` (multiple of 8) so the tiled MFMA
+// conv/wgrad path can consume it as a plain GEMM operand.
+template <typename T>
+__global__ void im2col_k(const T* __restrict__ x, T* __restrict__ out,
+                         int N, int H, int W, int C, int R, int S, int P,
+                         int Q, int sh, int sw, int ph, int pw, int ldout) {
+  const int kred_total = R * S * C;
+  const int64_t M = (int64_t)N * P * Q;
+  const int64_t total = M * (ldout / 8);
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int ug = (int)(i % (ldout / 8));
+    const int64_t m = i / (ldout / 8);
+    const int q = (int)(m % Q);
+    const int p = (int)((m / Q) % P);
+    const int n = (int)(m / ((int64_t)Q * P));
+    uint16_t e[8] = {0};
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int kred = ug * 8 + j;
+      if (kred < kred_total) {
+        const int r = kred / (S * C);
+        const int rem = kred % (S * C);
+        const int ss = rem / C;
+        const int c = rem % C;
+        const int hi = p * sh + r - ph;
+        const int wi = q * sw + ss - pw;
+        if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+          e[j] = *reinterpret_cast<const uint16_t*>(
+              x + (((int64_t)n * H + hi) * W + wi) * C + c);
+      }
+    }
+    *reinterpret_cast<uint4*>(out + m * ldout + ug * 8) =
+        *reinterpret_cast<const uint4*>(e);
+  }
+}
+
+// zero-pad the K x kred weight matrix into K x ldout
+template <typename T>
+__global__ void pad_cols_k(const T* __restrict__ w, T* __restrict__ out,
+                           int K, int kred, int ldout) {
+  const int64_t total = (int64_t)K * ldout;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int col = (int)(i % ldout);
+    const int64_t k = i / ldout;
+    out[i] = col < kred ? w[k * kred + col] : from_f32<T>(0.f);
+  }
+}
+
+// narrow K x ldout back to K x kred (wgrad output)
+template <typename T>
+__global__ void narrow_cols_k(const T* __restrict__ w, T* __restrict__ out,
+                              int K, int kred, int ldout) {
+  const int64_t total = (int64_t)K * kred;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int col = (int)(i % kred);
+    const int64_t k = i / kred;
+    out[i] = w[k * ldout + col];
+  }
+}
+
 inline int ew_grid(int64_t n, int per_thread = 8, int block = 256) {
   int64_t blocks = cdiv(n, (int64_t)block * per_thread);
   return (int)std::min<int64_t>(blocks, 2048);
@@ -207,6 +268,34 @@ void ddl_cast_f32_to(const float* src, void* dst, int64_t n, int dtype,
   int g = ew_grid(n, 4);
 #define RUN(T) cast_f32_k<T><<<g, 256, 0, s>>>(src, (T*)dst, n)
   switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_im2col(const void* x, void* out, int N, int H, int W, int C, int R,
+                int S, int P, int Q, int sh, int sw, int ph, int pw,
+                int ldout, int dtype, hipStream_t s) {
+  int64_t total = (int64_t)N * P * Q * (ldout / 8);
+  int g = (int)std::min<int64_t>(cdiv(total, 256), 4096);
+#define RUN(T)                                                                 \
+  im2col_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, N, H, W, C, R, S, P,    \
+                                Q, sh, sw, ph, pw, ldout)
+  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_pad_cols(const void* w, void* out, int K, int kred, int ldout,
+                  int dtype, hipStream_t s) {
+  int g = (int)std::min<int64_t>(cdiv((int64_t)K * ldout, 256), 1024);
+#define RUN(T) pad_cols_k<T><<<g, 256, 0, s>>>((const T*)w, (T*)out, K, kred, ldout)
+  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+void ddl_narrow_cols(const void* w, void* out, int K, int kred, int ldout,
+                     int dtype, hipStream_t s) {
+  int g = (int)std::min<int64_t>(cdiv((int64_t)K * kred, 256), 1024);
+#define RUN(T) narrow_cols_k<T><<<g, 256, 0, s>>>((const T*)w, (T*)out, K, kred, ldout)
+  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }
 
