@@ -54,11 +54,30 @@ Conv conv_geom(const at::Tensor& x, const at::Tensor& w, int sh, int sw,
 // For C % 8 != 0 (the 7x7 C=3 stem) materialise a padded im2col buffer and
 // run the tiled MFMA path as a plain GEMM — 2-3x the flat scalar-gather
 // kernel's speed for ~1 GB of extra (stream-rate) traffic.
+static at::Tensor& im2col_table(int R, int S, int C, int ld,
+                                const at::Device& dev) {
+  static std::map<std::tuple<int, int, int, int>, at::Tensor> cache;
+  auto key = std::make_tuple(R, S, C, ld);
+  auto it = cache.find(key);
+  if (it == cache.end()) {
+    auto t = at::empty({ld}, at::TensorOptions().dtype(at::kInt));
+    int* p = t.data_ptr<int>();
+    for (int k = 0; k < ld; ++k) {
+      int r = k / (S * C), rem = k % (S * C);
+      p[k] = (r << 20) | ((rem / C) << 10) | (rem % C);
+    }
+    it = cache.emplace(key, t.to(dev)).first;
+  }
+  return it->second;
+}
+
 static at::Tensor build_im2col(const at::Tensor& x, const Conv& g, int sh,
                                int sw, int ph, int pw, int ld) {
   auto col = at::empty({(int64_t)g.N * g.P * g.Q, ld}, x.options());
-  ddl_im2col(x.data_ptr(), col.data_ptr(), g.N, g.H, g.W, g.C, g.R, g.S,
-             g.P, g.Q, sh, sw, ph, pw, ld, dt_code(x), cur_stream());
+  auto& tbl = im2col_table(g.R, g.S, g.C, ld, x.device());
+  ddl_im2col(x.data_ptr(), col.data_ptr(), tbl.data_ptr<int>(), g.N, g.H,
+             g.W, g.C, g.R, g.S, g.P, g.Q, sh, sw, ph, pw, ld, dt_code(x),
+             cur_stream());
   return col;
 }
 

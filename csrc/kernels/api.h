@@ -110,9 +110,9 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
                     hipStream_t s);
 // im2col (small-C stem): x -> [N*P*Q, ldout] with (r,s,c) flattened cols
 // zero-padded to ldout (multiple of 8); plus K x kred <-> K x ldout pads.
-void ddl_im2col(const void* x, void* out, int N, int H, int W, int C, int R,
-                int S, int P, int Q, int sh, int sw, int ph, int pw,
-                int ldout, int dtype, hipStream_t s);
+void ddl_im2col(const void* x, void* out, const int* tbl, int N, int H,
+                int W, int C, int R, int S, int P, int Q, int sh, int sw,
+                int ph, int pw, int ldout, int dtype, hipStream_t s);
 void ddl_pad_cols(const void* w, void* out, int K, int kred, int ldout,
                   int dtype, hipStream_t s);
 void ddl_narrow_cols(const void* w, void* out, int K, int kred, int ldout,

@@ -22,9 +22,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   // no atomics (a chunks-deep atomicAdd chain per dw element serializes at
   // L2 like the BN-stats case), and the final sum is deterministic.
   dw_ws += (int64_t)blockIdx.z * slab_len;
-  __shared__ __align__(16) char smem[2 * TILE_BYTES];
-  char* ldsA = smem;                 // dy^T tile: [128 k][64 m]
-  char* ldsB = smem + TILE_BYTES;    // x^T tile:  [128 c|kred][64 m]
+  __shared__ __align__(16) char smem[4 * TILE_BYTES];  // A0 B0 A1 B1
+  char* const lA[2] = {smem, smem + 2 * TILE_BYTES};   // dy^T: [128 k][64 m]
+  char* const lB[2] = {smem + TILE_BYTES, smem + 3 * TILE_BYTES};  // x^T
 
   const int kred_total = (MODE == FLAT) ? d.R * d.S * d.C : d.C;
   const int ntc = (kred_total + BN - 1) / BN;
@@ -47,94 +47,93 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
 
   const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
   const int64_t m_end = min(M, m_begin + chunk_rows);
+  const int64_t ntiles = (m_end > m_begin) ? (m_end - m_begin + BK - 1) / BK : 0;
 
-  for (int64_t m0 = m_begin; m0 < m_end; m0 += BK) {
-    // ---- stage dy^T: load 4 m-rows of 8 k, write LDS[k][m] b64 -----------
-    {
-      uint4 v[4];
-      const int k0g = tile_k + su * 8;
+  uint4 va[4], vb[4];
+  auto load_tile = [&](int64_t ti) {
+    const int64_t m0 = m_begin + ti * BK;
+    const int k0g = tile_k + su * 8;
 #pragma unroll
-      for (int mm = 0; mm < 4; ++mm) {
-        const int64_t m = m0 + sm4 + mm;
-        v[mm] = zero16();
-        if (m < m_end && k0g + 8 <= d.K)
-          v[mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
-      }
-      const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
-      const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
-      const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
-      const uint16_t* e3 = reinterpret_cast<const uint16_t*>(&v[3]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
-                      ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
-        *reinterpret_cast<uint64_t*>(ldsA + swz_off(su * 8 + j, sm4 * 2)) = pk;
-      }
+    for (int mm = 0; mm < 4; ++mm) {
+      const int64_t m = m0 + sm4 + mm;
+      va[mm] = zero16();
+      if (m < m_end && k0g + 8 <= d.K)
+        va[mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
     }
-    // ---- stage x^T: im2col gather, same transposed write -----------------
-    {
-      uint4 v[4];
 #pragma unroll
-      for (int mm = 0; mm < 4; ++mm) {
-        const int64_t m = m0 + sm4 + mm;
-        v[mm] = zero16();
-        uint16_t* e = reinterpret_cast<uint16_t*>(&v[mm]);
-        if (m < m_end) {
-          const int q = (int)(m % d.Q);
-          const int p = (int)((m / d.Q) % d.P);
-          const int n = (int)(m / ((int64_t)d.Q * d.P));
-          if (MODE == TILED) {
-            const int hi = p * d.sh + r - d.ph;
-            const int wi = q * d.sw + s - d.pw;
-            const int c0 = tile_c + su * 8;
-            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
-              v[mm] = *reinterpret_cast<const uint4*>(
-                  x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
-          } else {
+    for (int mm = 0; mm < 4; ++mm) {
+      const int64_t m = m0 + sm4 + mm;
+      vb[mm] = zero16();
+      uint16_t* e = reinterpret_cast<uint16_t*>(&vb[mm]);
+      if (m < m_end) {
+        const int q = (int)(m % d.Q);
+        const int p = (int)((m / d.Q) % d.P);
+        const int n = (int)(m / ((int64_t)d.Q * d.P));
+        if (MODE == TILED) {
+          const int hi = p * d.sh + r - d.ph;
+          const int wi = q * d.sw + s - d.pw;
+          const int c0 = tile_c + su * 8;
+          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
+            vb[mm] = *reinterpret_cast<const uint4*>(
+                x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
+        } else {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int kred = tile_c + su * 8 + j;
-              if (kred < kred_total) {
-                const int rr = kred / (d.S * d.C);
-                const int rem = kred % (d.S * d.C);
-                const int ss = rem / d.C;
-                const int cc = rem % d.C;
-                const int hi = p * d.sh + rr - d.ph;
-                const int wi = q * d.sw + ss - d.pw;
-                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                  e[j] = *reinterpret_cast<const uint16_t*>(
-                      x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
-              }
+          for (int j = 0; j < 8; ++j) {
+            const int kred = tile_c + su * 8 + j;
+            if (kred < kred_total) {
+              const int rr = kred / (d.S * d.C);
+              const int rem = kred % (d.S * d.C);
+              const int ss = rem / d.C;
+              const int cc = rem % d.C;
+              const int hi = p * d.sh + rr - d.ph;
+              const int wi = q * d.sw + ss - d.pw;
+              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                e[j] = *reinterpret_cast<const uint16_t*>(
+                    x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
             }
           }
         }
       }
-      const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
-      const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
-      const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
-      const uint16_t* e3 = reinterpret_cast<const uint16_t*>(&v[3]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
-                      ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
-        *reinterpret_cast<uint64_t*>(ldsB + swz_off(su * 8 + j, sm4 * 2)) = pk;
-      }
     }
-    __syncthreads();
+  };
 
-    const int frow = lane & 15;
-    const int fcolb = (lane >> 4) * 16;
+  auto write_tile = [&](char* la, char* lb) {
+    const uint16_t* a0 = reinterpret_cast<const uint16_t*>(&va[0]);
+    const uint16_t* a1 = reinterpret_cast<const uint16_t*>(&va[1]);
+    const uint16_t* a2 = reinterpret_cast<const uint16_t*>(&va[2]);
+    const uint16_t* a3 = reinterpret_cast<const uint16_t*>(&va[3]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      uint64_t pk = (uint64_t)a0[j] | ((uint64_t)a1[j] << 16) |
+                    ((uint64_t)a2[j] << 32) | ((uint64_t)a3[j] << 48);
+      *reinterpret_cast<uint64_t*>(la + swz_off(su * 8 + j, sm4 * 2)) = pk;
+    }
+    const uint16_t* b0 = reinterpret_cast<const uint16_t*>(&vb[0]);
+    const uint16_t* b1 = reinterpret_cast<const uint16_t*>(&vb[1]);
+    const uint16_t* b2 = reinterpret_cast<const uint16_t*>(&vb[2]);
+    const uint16_t* b3 = reinterpret_cast<const uint16_t*>(&vb[3]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      uint64_t pk = (uint64_t)b0[j] | ((uint64_t)b1[j] << 16) |
+                    ((uint64_t)b2[j] << 32) | ((uint64_t)b3[j] << 48);
+      *reinterpret_cast<uint64_t*>(lb + swz_off(su * 8 + j, sm4 * 2)) = pk;
+    }
+  };
+
+  const int frow = lane & 15;
+  const int fcolb = (lane >> 4) * 16;
+  auto mfma_tile = [&](const char* la, const char* lb) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       typename MfmaFrag<T>::type a[4], b[4];
 #pragma unroll
       for (int mr = 0; mr < 4; ++mr) {
-        uint4 v = lds_read16(ldsA, wr + mr * 16 + frow, kk * 64 + fcolb);
+        uint4 v = lds_read16(la, wr + mr * 16 + frow, kk * 64 + fcolb);
         a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
       }
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr) {
-        uint4 v = lds_read16(ldsB, wc + nr * 16 + frow, kk * 64 + fcolb);
+        uint4 v = lds_read16(lb, wc + nr * 16 + frow, kk * 64 + fcolb);
         b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
       }
 #pragma unroll
@@ -143,7 +142,20 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
         for (int nr = 0; nr < 4; ++nr)
           acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
     }
+  };
+
+  if (ntiles > 0) {
+    load_tile(0);
+    write_tile(lA[0], lB[0]);
     __syncthreads();
+    for (int64_t t = 0; t < ntiles; ++t) {
+      const bool more = (t + 1) < ntiles;
+      const int cur = (int)(t & 1);
+      if (more) load_tile(t + 1);
+      mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
+      if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: plain dense stores into this chunk's fp32 slab ----------
@@ -218,9 +230,9 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   pl.ntc = (int)cdiv(pl.kred_total, BN);
   pl.rs_total = flat ? 1 : R * S;
   int64_t chunks = std::max<int64_t>(
-      1, 768 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
+      1, 1024 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
   chunks = std::min<int64_t>(chunks, cdiv(M, BK));
-  chunks = std::min<int64_t>(chunks, 128);  // bound slab traffic
+  chunks = std::min<int64_t>(chunks, 512);  // bound slab traffic
   // cap slab memory at ~256 MB
   const int64_t slab = (int64_t)K * R * S * C * 4;
   if (slab > 0) chunks = std::min<int64_t>(chunks, std::max<int64_t>(1, (256LL << 20) / slab));

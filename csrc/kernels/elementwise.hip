@@ -149,11 +149,13 @@ __global__ void transpose_kc_k(const T* w, T* w_t, int K, int RS, int C) {
 // im2col for small-C stems (conv1: 7x7 s2 C=3): rows m=(n,p,q), cols
 // kred=(r,s,c) zero-padded to `ldout` (multiple of 8) so the tiled MFMA
 // conv/wgrad path can consume it as a plain GEMM operand.
+// tbl[kred] = (r<<20)|(s<<10)|c precomputed host-side (integer divides in
+// the inner gather cost ~8 divides/16B and dominated the first version).
 template <typename T>
 __global__ void im2col_k(const T* __restrict__ x, T* __restrict__ out,
-                         int N, int H, int W, int C, int R, int S, int P,
-                         int Q, int sh, int sw, int ph, int pw, int ldout) {
-  const int kred_total = R * S * C;
+                         const int* __restrict__ tbl, int N, int H, int W,
+                         int C, int P, int Q, int sh, int sw, int ph, int pw,
+                         int kred_total, int ldout) {
   const int64_t M = (int64_t)N * P * Q;
   const int64_t total = M * (ldout / 8);
   for (int64_t i = global_tid(); i < total; i += grid_stride()) {
@@ -162,20 +164,18 @@ __global__ void im2col_k(const T* __restrict__ x, T* __restrict__ out,
     const int q = (int)(m % Q);
     const int p = (int)((m / Q) % P);
     const int n = (int)(m / ((int64_t)Q * P));
+    const int h0 = p * sh - ph, w0 = q * sw - pw;
     uint16_t e[8] = {0};
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int kred = ug * 8 + j;
       if (kred < kred_total) {
-        const int r = kred / (S * C);
-        const int rem = kred % (S * C);
-        const int ss = rem / C;
-        const int c = rem % C;
-        const int hi = p * sh + r - ph;
-        const int wi = q * sw + ss - pw;
+        const int t = tbl[kred];
+        const int hi = h0 + (t >> 20);
+        const int wi = w0 + ((t >> 10) & 1023);
         if (hi >= 0 && hi < H && wi >= 0 && wi < W)
           e[j] = *reinterpret_cast<const uint16_t*>(
-              x + (((int64_t)n * H + hi) * W + wi) * C + c);
+              x + (((int64_t)n * H + hi) * W + wi) * C + (t & 1023));
       }
     }
     *reinterpret_cast<uint4*>(out + m * ldout + ug * 8) =
@@ -271,14 +271,15 @@ void ddl_cast_f32_to(const float* src, void* dst, int64_t n, int dtype,
 #undef RUN
 }
 
-void ddl_im2col(const void* x, void* out, int N, int H, int W, int C, int R,
-                int S, int P, int Q, int sh, int sw, int ph, int pw,
-                int ldout, int dtype, hipStream_t s) {
+void ddl_im2col(const void* x, void* out, const int* tbl, int N, int H,
+                int W, int C, int R, int S, int P, int Q, int sh, int sw,
+                int ph, int pw, int ldout, int dtype, hipStream_t s) {
   int64_t total = (int64_t)N * P * Q * (ldout / 8);
-  int g = (int)std::min<int64_t>(cdiv(total, 256), 4096);
+  int g = (int)std::min<int64_t>(cdiv(total, 256), 8192);
+  int kred_total = R * S * C;
 #define RUN(T)                                                                 \
-  im2col_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, N, H, W, C, R, S, P,    \
-                                Q, sh, sw, ph, pw, ldout)
+  im2col_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, tbl, N, H, W, C, P,     \
+                                Q, sh, sw, ph, pw, kred_total, ldout)
   switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }
