@@ -4,6 +4,8 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <c10/util/Optional.h>
+#include <map>
+#include <tuple>
 
 #include "kernels/api.h"
 
@@ -111,6 +113,49 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w,
                g.C, g.K, g.R, g.S, g.P, g.Q, sh, sw, ph, pw, relu, dt,
                cur_stream());
   return y;
+}
+
+// small-C (stem) variant returning the im2col buffer so backward reuses it
+std::tuple<at::Tensor, at::Tensor> conv_fwd_smallc(
+    at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias, int64_t sh,
+    int64_t sw, int64_t ph, int64_t pw, bool relu) {
+  check_cl(x, "x"); check_cl(w, "w");
+  Conv g = conv_geom(x, w, sh, sw, ph, pw);
+  int dt = dt_code(x);
+  const int kred = g.R * g.S * g.C;
+  const int ld = (kred + 7) & ~7;
+  auto col = build_im2col(x, g, sh, sw, ph, pw, ld);
+  auto wp = at::empty({g.K, ld}, w.options());
+  ddl_pad_cols(w.data_ptr(), wp.data_ptr(), g.K, kred, ld, dt, cur_stream());
+  auto y = empty_nhwc(g.N, g.K, g.P, g.Q, x.options());
+  const void* bptr = (bias.has_value() && bias->defined())
+                         ? bias->data_ptr() : nullptr;
+  ddl_conv_fwd(col.data_ptr(), wp.data_ptr(), bptr, y.data_ptr(), g.N, g.P,
+               g.Q, ld, g.K, 1, 1, g.P, g.Q, 1, 1, 0, 0, relu, dt,
+               cur_stream());
+  return {y, col};
+}
+
+at::Tensor conv_wgrad_smallc(at::Tensor dy, at::Tensor col,
+                             std::vector<int64_t> w_shape,
+                             std::vector<int64_t> x_shape) {
+  check_cl(dy, "dy");
+  int N = x_shape[0], C = x_shape[1];
+  int K = w_shape[0], R = w_shape[2], S = w_shape[3];
+  int P = dy.size(2), Q = dy.size(3);
+  int dt = dt_code(dy);
+  const int kred = R * S * C;
+  const int ld = (int)col.size(1);
+  auto ws = at::empty({ddl_conv_wgrad_ws_elems(N, ld, K, 1, 1, P, Q)},
+                      dy.options().dtype(at::kFloat));
+  auto dwp = at::empty({K, ld}, dy.options());
+  ddl_conv_wgrad(dy.data_ptr(), col.data_ptr(), ws.data_ptr<float>(),
+                 dwp.data_ptr(), N, P, Q, ld, K, 1, 1, P, Q, 1, 1, 0, 0,
+                 dt, cur_stream());
+  auto dw = empty_nhwc(K, C, R, S, dy.options());
+  ddl_narrow_cols(dwp.data_ptr(), dw.data_ptr(), K, kred, ld, dt,
+                  cur_stream());
+  return dw;
 }
 
 at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
@@ -392,6 +437,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);
+  m.def("conv_fwd_smallc", &conv_fwd_smallc);
+  m.def("conv_wgrad_smallc", &conv_wgrad_smallc);
   m.def("bn_relu_fwd", &bn_relu_fwd);
   m.def("bn_relu_infer", &bn_relu_infer);
   m.def("bn_relu_bwd", &bn_relu_bwd);

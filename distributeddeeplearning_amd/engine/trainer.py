@@ -4,6 +4,7 @@ sync point is ``DistributedOptimizer.synchronize()`` inside ``step``."""
 from __future__ import annotations
 
 import math
+import os
 import time
 from typing import Iterator, Optional, Tuple
 
@@ -76,6 +77,19 @@ class Trainer:
         self.global_step = 0
         self.epoch = 0
 
+        # Chrome-trace timeline (HOROVOD_TIMELINE equivalent): engine-level
+        # spans from Python, per-collective spans from the native comm engine.
+        tl_path = cfg.timeline or os.environ.get("DDL_TIMELINE")
+        self.timeline = None
+        if tl_path:
+            from ..utils.timeline import Timeline
+            self.timeline = Timeline(f"{tl_path}.rank{self.rank}.json", self.rank)
+            self.opt.set_timeline(self.timeline)
+            backend = api._backend
+            if hasattr(backend, "engine"):
+                backend.engine.enable_timeline(
+                    f"{tl_path}.rank{self.rank}.comm.json")
+
         if cfg.resume:
             payload = ckpt.load(cfg.resume, self.model, self.opt, self.scaler,
                                 device=self.device)
@@ -111,10 +125,19 @@ class Trainer:
         return self._loader
 
     def train_step(self, x, y) -> torch.Tensor:
+        tl = self.timeline
         lr = self.lr_sched.step(self.global_step)
+        if tl:
+            tl.begin("forward")
         logits = self.model(x)
         loss = DF.softmax_cross_entropy(logits, y, self.cfg.label_smoothing)
+        if tl:
+            tl.end("forward")
+            tl.begin("backward+allreduce")
         self.scaler.scale_loss(loss).backward()
+        if tl:
+            tl.end("backward+allreduce")
+            tl.begin("step")
         if self.scaler.enabled:
             self.opt.synchronize()
             flats = [b.flat for b in self.opt.buckets.buckets]
@@ -125,6 +148,10 @@ class Trainer:
         else:
             self.opt.step()
         self.opt.zero_grad()
+        if tl:
+            tl.end("step")
+            if self.global_step % 50 == 0:
+                tl.save()
         self.global_step += 1
         return loss
 

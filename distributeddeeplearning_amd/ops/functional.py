@@ -42,12 +42,20 @@ class _Conv2dFn(torch.autograd.Function):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         weight = to_channels_last(weight)
-        y = ext.conv_fwd(x, weight, bias if bias is not None else None,
-                         stride[0], stride[1], padding[0], padding[1], relu)
-        if relu:
-            ctx.save_for_backward(x, weight, y)
+        ctx.small_c = x.shape[1] % 8 != 0
+        if ctx.small_c:
+            # stem path: keep the im2col buffer for wgrad reuse
+            y, col = ext.conv_fwd_smallc(
+                x, weight, bias if bias is not None else None,
+                stride[0], stride[1], padding[0], padding[1], relu)
+            saved = [x, weight, col]
         else:
-            ctx.save_for_backward(x, weight)
+            y = ext.conv_fwd(x, weight, bias if bias is not None else None,
+                             stride[0], stride[1], padding[0], padding[1], relu)
+            saved = [x, weight]
+        if relu:
+            saved.append(y)
+        ctx.save_for_backward(*saved)
         ctx.stride = stride
         ctx.padding = padding
         ctx.has_bias = bias is not None
@@ -57,21 +65,34 @@ class _Conv2dFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = load_kernels_ext()
+        saved = list(ctx.saved_tensors)
         if ctx.relu:
-            x, weight, y = ctx.saved_tensors
+            y = saved.pop()
+        if ctx.small_c:
+            x, weight, col = saved
+        else:
+            x, weight = saved
+            col = None
+        if ctx.relu:
             dy = ext.relu_bwd(to_channels_last(dy), y)
         else:
-            x, weight = ctx.saved_tensors
             dy = to_channels_last(dy)
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
+            if ctx.small_c:
+                raise RuntimeError("dgrad for C%8!=0 (stem) is not needed "
+                                   "(input has no grad) and not implemented")
             dx = ext.conv_dgrad(dy, weight, list(x.shape),
                                 ctx.stride[0], ctx.stride[1],
                                 ctx.padding[0], ctx.padding[1])
         if ctx.needs_input_grad[1]:
-            dw = ext.conv_wgrad(dy, x, list(weight.shape),
-                                ctx.stride[0], ctx.stride[1],
-                                ctx.padding[0], ctx.padding[1])
+            if ctx.small_c:
+                dw = ext.conv_wgrad_smallc(dy, col, list(weight.shape),
+                                           list(x.shape))
+            else:
+                dw = ext.conv_wgrad(dy, x, list(weight.shape),
+                                    ctx.stride[0], ctx.stride[1],
+                                    ctx.padding[0], ctx.padding[1])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.sum(dim=(0, 2, 3))
         return dx, dw, db, None, None, None

@@ -1,0 +1,123 @@
+"""Auxiliary subsystems: timeline, loss scaler, config presets, imagenet
+folder pipeline, report tooling, single-process API."""
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+import torch
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_timeline_writer(tmp_path):
+    from distributeddeeplearning_amd.utils.timeline import Timeline
+    p = str(tmp_path / "tl.json")
+    tl = Timeline(p, rank=0)
+    tl.begin("forward")
+    tl.end("forward")
+    tl.mark("bucket0/launch")
+    tl.save()
+    data = json.load(open(p))
+    names = [e["name"] for e in data["traceEvents"]]
+    assert "forward" in names and "bucket0/launch" in names
+    assert any(e.get("ph") == "X" for e in data["traceEvents"])
+
+
+def test_loss_scaler_dynamics():
+    from distributeddeeplearning_amd.engine.loss_scaler import DynamicLossScaler
+    s = DynamicLossScaler(init_scale=1024.0, growth_interval=2)
+    good = [torch.ones(4)]
+    bad = [torch.tensor([1.0, float("inf")])]
+    assert s.check_and_update(good)
+    assert s.check_and_update(good)
+    assert s.scale == 2048.0  # grew after 2 good steps
+    assert not s.check_and_update(bad)
+    assert s.scale == 1024.0  # backed off
+    assert s.skipped_steps == 1
+
+
+def test_config_presets_cover_baseline():
+    from distributeddeeplearning_amd.config import PRESETS
+    assert set(PRESETS) == {"cpu-plumbing", "resnet50-bf16",
+                            "resnet50-fp16-lars-b8k", "resnet152-bf16",
+                            "vgg16-bf16"}
+    b8k = PRESETS["resnet50-fp16-lars-b8k"]
+    assert b8k.lars and b8k.dtype == "fp16"
+    assert b8k.batch_size * 8 == 8192  # global batch 8k at DP=8
+
+
+def test_config_from_args_preset_override():
+    from distributeddeeplearning_amd.config import TrainConfig
+    cfg = TrainConfig.from_args(["--preset", "vgg16-bf16", "--batch-size", "64"])
+    assert cfg.model == "vgg16"
+    assert cfg.batch_size == 64  # CLI wins over preset
+
+
+def test_imagenet_folder_pipeline(tmp_path):
+    from PIL import Image
+    for cls in ["beagle", "tabby"]:
+        d = tmp_path / "train" / cls
+        d.mkdir(parents=True)
+        for i in range(4):
+            arr = (np.random.rand(40, 52, 3) * 255).astype("uint8")
+            Image.fromarray(arr).save(d / f"img{i}.jpg")
+    from distributeddeeplearning_amd.data.imagenet import (
+        ImageFolderDataset, ShardedSampler, imagenet_loader)
+    ds = ImageFolderDataset(str(tmp_path / "train"), image_size=32)
+    assert len(ds) == 8
+    x, y = ds[0]
+    assert x.shape == (3, 32, 32) and y in (0, 1)
+    smp = ShardedSampler(len(ds), rank=1, world=2)
+    assert len(list(iter(smp))) == 4
+    it = imagenet_loader(str(tmp_path), batch_size=2, image_size=32,
+                         dtype=torch.float32, device=None, rank=0, world=2,
+                         channels_last=False, num_workers=0)
+    xb, yb = next(it)
+    assert xb.shape == (2, 3, 32, 32)
+
+
+def test_report_scaling(tmp_path, capsys):
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    import report
+    lines = [
+        {"metric": "images/sec (whole node) ResNet-50 synthetic ImageNet",
+         "value": 5000.0, "n_gpus": 1},
+        {"metric": "images/sec (whole node) ResNet-50 synthetic ImageNet",
+         "value": 38000.0, "n_gpus": 8},
+    ]
+    p = tmp_path / "b.json"
+    p.write_text("\n".join(json.dumps(l) for l in lines))
+    report.scaling_table([str(p)])
+    out = capsys.readouterr().out
+    assert "95.0%" in out
+
+
+def test_single_process_api_collectives():
+    from distributeddeeplearning_amd.parallel import api
+    api.init("single")
+    t = torch.arange(4.0)
+    assert torch.equal(api.allreduce(t), t)
+    assert torch.equal(api.allgather(t), t)
+    h = api.allreduce_async_(t.clone())
+    api.synchronize(h)
+    opt = torch.optim.SGD([torch.nn.Parameter(torch.ones(2))], lr=0.1)
+    api.broadcast_optimizer_state(opt, 0)  # no-op at world 1
+    api.broadcast_parameters({"w": t}, 0)
+
+
+def test_launcher_spawns_and_fails_fast(tmp_path):
+    script = tmp_path / "child.py"
+    script.write_text(
+        "import os, sys\n"
+        "rank = int(os.environ['RANK'])\n"
+        "print('rank', rank, 'of', os.environ['WORLD_SIZE'])\n"
+        "sys.exit(3 if rank == 1 else 0)\n")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "launch.py"),
+         "--nproc", "2", str(script)],
+        capture_output=True, text=True, timeout=60)
+    assert r.returncode == 3
+    assert "rank 0 of 2" in r.stdout
