@@ -21,14 +21,20 @@ namespace {
 
 enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2 };
 
-template <typename T, int MODE>
+// DBUF=true: double-buffered LDS (64 KiB, 2 blocks/CU) with loads issued
+// under the MFMAs — best for deep reduce loops.  DBUF=false: single buffer
+// (32 KiB, ~4 blocks/CU) with issue-early loads and two barriers per tile —
+// the shallow-reduce 56x56/64-channel layers are LATENCY-bound (their whole
+// K-loop is 1-9 tiles of ~600 MFMA cycles), so occupancy beats pipelining.
+template <typename T, int MODE, bool DBUF>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
     int Nc, int relu) {
-  __shared__ __align__(16) char smem[4 * TILE_BYTES];  // A0 B0 A1 B1
-  char* const lA[2] = {smem, smem + 2 * TILE_BYTES};
-  char* const lB[2] = {smem + TILE_BYTES, smem + 3 * TILE_BYTES};
+  __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
+  char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
+  char* const lB[2] = {smem + TILE_BYTES,
+                       DBUF ? smem + 3 * TILE_BYTES : smem + TILE_BYTES};
 
   const int nt_n = (Nc + BN - 1) / BN;
   const unsigned bid = xcd_swizzle(blockIdx.x, gridDim.x);
@@ -196,19 +202,31 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   };
 
-  // ---- pipelined main loop: one barrier per tile -------------------------
-  load_tile(0);
-  write_tile(lA[0], lB[0]);
-  __syncthreads();
-  for (int t = 0; t < total_tiles; ++t) {
-    const int cur = t & 1;
-    const bool more = (t + 1) < total_tiles;
-    if (more) load_tile(t + 1);                 // global loads in flight...
-    char* la = cur ? lA[1] : lA[0];
-    char* lb = cur ? lB[1] : lB[0];
-    mfma_tile(la, lb);                          // ...hidden under the MFMAs
-    if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+  if (DBUF) {
+    // ---- pipelined main loop: one barrier per tile -----------------------
+    load_tile(0);
+    write_tile(lA[0], lB[0]);
     __syncthreads();
+    for (int t = 0; t < total_tiles; ++t) {
+      const int cur = t & 1;
+      const bool more = (t + 1) < total_tiles;
+      if (more) load_tile(t + 1);               // global loads in flight...
+      char* la = cur ? lA[1] : lA[0];
+      char* lb = cur ? lB[1] : lB[0];
+      mfma_tile(la, lb);                        // ...hidden under the MFMAs
+      if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      __syncthreads();
+    }
+  } else {
+    // ---- single buffer, issue-early loads, two barriers per tile ---------
+    load_tile(0);
+    for (int t = 0; t < total_tiles; ++t) {
+      write_tile(lA[0], lB[0]);
+      __syncthreads();
+      if (t + 1 < total_tiles) load_tile(t + 1);  // overlap with MFMAs
+      mfma_tile(lA[0], lB[0]);
+      __syncthreads();
+    }
   }
 
   // ---- epilogue: D[m][n], m = (lane>>4)*4 + reg, n = lane&15 --------------
@@ -233,6 +251,13 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   }
 }
 
+bool conv_use_dbuf(int cred, int rs_total) {
+  const char* env = getenv("DDL_CONV_DBUF");
+  if (env) return env[0] == '1';
+  // shallow reduce loops are latency-bound: favour occupancy
+  return (int64_t)cred * rs_total > 9 * 64;
+}
+
 }  // namespace
 
 extern "C" {
@@ -246,11 +271,15 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
   const int Nc = K;
   const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
   const bool flat = (C % 8) != 0;
-#define RUN(T, MODE)                                                           \
-  conv_igemm_k<T, MODE><<<(unsigned)grid, 256, 0, s>>>(                        \
+  const bool dbuf = conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
+#define RUN(T, MODE, DB)                                                       \
+  conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
       (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu)
-  if (dtype == 1) { if (flat) RUN(bf16, FWD_FLAT); else RUN(bf16, FWD); }
-  else if (dtype == 2) { if (flat) RUN(f16, FWD_FLAT); else RUN(f16, FWD); }
+#define SEL(T, MODE)                                                           \
+  if (dbuf) RUN(T, MODE, true); else RUN(T, MODE, false)
+  if (dtype == 1) { if (flat) { SEL(bf16, FWD_FLAT); } else { SEL(bf16, FWD); } }
+  else if (dtype == 2) { if (flat) { SEL(f16, FWD_FLAT); } else { SEL(f16, FWD); } }
+#undef SEL
 #undef RUN
 }
 
@@ -261,11 +290,12 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
   const int64_t M = (int64_t)N * H * W;
   const int Nc = C;
   const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
-#define RUN(T)                                                                 \
-  conv_igemm_k<T, DGRAD><<<(unsigned)grid, 256, 0, s>>>(                       \
+  const bool dbuf = conv_use_dbuf(K, R * S);
+#define RUN(T, DB)                                                             \
+  conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
       (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0)
-  if (dtype == 1) RUN(bf16);
-  else if (dtype == 2) RUN(f16);
+  if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
+  else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
 #undef RUN
 }
 

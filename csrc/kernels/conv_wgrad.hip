@@ -13,7 +13,9 @@ namespace {
 
 enum WMode { TILED = 0, FLAT = 1 };
 
-template <typename T, int MODE>
+// SIMPLE = 1x1/stride-1/no-pad conv: the x operand is just row m of the
+// input (no im2col decode; the per-tile q/p/n divides were a hot cost).
+template <typename T, int MODE, bool SIMPLE>
 __global__ __launch_bounds__(256) void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
     float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows,
@@ -65,7 +67,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
       const int64_t m = m0 + sm4 + mm;
       vb[mm] = zero16();
       uint16_t* e = reinterpret_cast<uint16_t*>(&vb[mm]);
-      if (m < m_end) {
+      if (SIMPLE) {
+        const int c0 = tile_c + su * 8;
+        if (m < m_end && c0 + 8 <= d.C)
+          vb[mm] = *reinterpret_cast<const uint4*>(x + m * d.C + c0);
+      } else if (m < m_end) {
         const int q = (int)(m % d.Q);
         const int p = (int)((m / d.Q) % d.P);
         const int n = (int)(m / ((int64_t)d.Q * d.P));
@@ -262,12 +268,20 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   const int64_t slab_len = (int64_t)K * R * S * C;
   dim3 grid((unsigned)(pl.ntk * pl.ntc), (unsigned)pl.rs_total,
             (unsigned)pl.chunks);
-#define RUN(T, MODE)                                                           \
-  conv_wgrad_k<T, MODE><<<grid, 256, 0, s>>>((const T*)dy, (const T*)x,        \
-                                             dw_ws, d, M, pl.chunk_rows,       \
-                                             slab_len)
-  if (dtype == 1) { if (flat) RUN(bf16, FLAT); else RUN(bf16, TILED); }
-  else if (dtype == 2) { if (flat) RUN(f16, FLAT); else RUN(f16, TILED); }
+  const bool simple = (R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
+                       pw == 0 && !flat);
+#define RUN(T, MODE, SIMP)                                                     \
+  conv_wgrad_k<T, MODE, SIMP><<<grid, 256, 0, s>>>(                            \
+      (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len)
+  if (dtype == 1) {
+    if (flat) RUN(bf16, FLAT, false);
+    else if (simple) RUN(bf16, TILED, true);
+    else RUN(bf16, TILED, false);
+  } else if (dtype == 2) {
+    if (flat) RUN(f16, FLAT, false);
+    else if (simple) RUN(f16, TILED, true);
+    else RUN(f16, TILED, false);
+  }
 #undef RUN
   int g = (int)cdiv(slab_len, 256);
 #define RED(T)                                                                 \
