@@ -69,6 +69,14 @@ class TrainConfig:
         return torch.device(self.device)
 
     @classmethod
+    def from_yaml(cls, path: str) -> "TrainConfig":
+        import yaml
+        with open(path) as f:
+            d = yaml.safe_load(f)
+        d = {k: tuple(v) if isinstance(v, list) else v for k, v in d.items()}
+        return cls(**d)
+
+    @classmethod
     def from_args(cls, argv=None, **overrides):
         parser = argparse.ArgumentParser(description="distributeddeeplearning_amd trainer")
         defaults = cls(**overrides)
@@ -86,7 +94,17 @@ class TrainConfig:
                 parser.add_argument(f"--{name}", type=typ, default=default)
         parser.add_argument("--preset", type=str, default=None,
                             help=f"named preset: {sorted(PRESETS)}")
+        parser.add_argument("--config", type=str, default=None,
+                            help="YAML config file (see configs/)")
         ns = parser.parse_args(argv)
+        if ns.config:
+            base = dataclasses.asdict(cls.from_yaml(ns.config))
+            given = {a.dest for a in parser._actions
+                     if getattr(ns, a.dest, None) != a.default
+                     and a.dest not in ("preset", "config")}
+            for k, v in base.items():
+                if k not in given:
+                    setattr(ns, k, v)
         if ns.preset:
             base = dataclasses.asdict(PRESETS[ns.preset])
             # CLI flags explicitly given override preset values
