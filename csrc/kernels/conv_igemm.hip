@@ -229,24 +229,41 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
 
-  // ---- epilogue: D[m][n], m = (lane>>4)*4 + reg, n = lane&15 --------------
+  // ---- epilogue through LDS: repack D fragments ([4 consecutive m] x
+  // [1 n] per lane) into a [128][128] element tile, then store coalesced
+  // 16B rows.  Direct fragment stores are 64 scalar 2B globals per lane —
+  // issue-bound (guide T21 diagnostic) and dominant for shallow layers.
+  __syncthreads();  // done with the operand tiles; reuse smem
+  T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
 #pragma unroll
   for (int nr = 0; nr < 4; ++nr) {
-    const int n = tile_n + wc + nr * 16 + en;
-    if (n >= Nc) continue;
-    const float bv = (MODE != DGRAD && bias != nullptr) ? to_f32(bias[n]) : 0.f;
+    const int n_loc = wc + nr * 16 + en;
+    const float bv = (MODE != DGRAD && bias != nullptr && tile_n + n_loc < Nc)
+                         ? to_f32(bias[tile_n + n_loc]) : 0.f;
 #pragma unroll
     for (int mr = 0; mr < 4; ++mr) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int64_t m = tile_m + wr + mr * 16 + em + reg;
-        if (m >= M) continue;
         float f = acc[mr][nr][reg] + bv;
         if (relu) f = fmaxf(f, 0.f);
-        out[m * Nc + n] = from_f32<T>(f);
+        etile[(wr + mr * 16 + em + reg) * BN + n_loc] = from_f32<T>(f);
       }
+    }
+  }
+  __syncthreads();
+  // 256 threads x 8 iterations of 16B: row-major sweep of the 128x128 tile
+  const int nvalid = Nc - tile_n;  // columns of this tile in range
+#pragma unroll
+  for (int it = 0; it < 8; ++it) {
+    const int unit = threadIdx.x + it * 256;   // 16B unit index (2048 total)
+    const int row = unit >> 4;                 // BN*2/16 = 16 units per row
+    const int ub = (unit & 15) * 8;            // first element in row
+    const int64_t m = tile_m + row;
+    if (m < M && ub < nvalid) {
+      *reinterpret_cast<uint4*>(out + m * Nc + tile_n + ub) =
+          *reinterpret_cast<const uint4*>(etile + row * BN + ub);
     }
   }
 }

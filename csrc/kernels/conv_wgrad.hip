@@ -164,7 +164,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
     }
   }
 
-  // ---- epilogue: plain dense stores into this chunk's fp32 slab ----------
+  // ---- epilogue: repack through LDS (fp32 [128][128] = the whole 64 KiB
+  // smem), then coalesced 16B stores into this chunk's fp32 slab ----------
+  __syncthreads();
+  float* etile = reinterpret_cast<float*>(smem);
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
   const int rsc = d.R * d.S * d.C;
@@ -172,16 +175,31 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   for (int mr = 0; mr < 4; ++mr) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      const int k = tile_k + wr + mr * 16 + em + reg;
-      if (k >= d.K) continue;
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr) {
-        const int cidx = tile_c + wc + nr * 16 + en;
-        if (cidx >= kred_total) continue;
-        const int64_t o = (MODE == FLAT)
-            ? (int64_t)k * rsc + cidx
-            : (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
-        dw_ws[o] = acc[mr][nr][reg];
+        etile[(wr + mr * 16 + em + reg) * BN + wc + nr * 16 + en] =
+            acc[mr][nr][reg];
+      }
+    }
+  }
+  __syncthreads();
+  const int nvalid = kred_total - tile_c;
+#pragma unroll
+  for (int it = 0; it < 16; ++it) {
+    const int unit = threadIdx.x + it * 256;   // 16B = 4 floats; 4096 units
+    const int row = unit >> 5;                 // 32 units per 128-f32 row
+    const int ub = (unit & 31) * 4;
+    const int k = tile_k + row;
+    if (k < d.K && ub < nvalid) {
+      const int cidx = tile_c + ub;
+      if (MODE == FLAT) {
+        // flat slab rows (R*S*C, e.g. 147) are not 16B-aligned: scalar
+        for (int j = 0; j < 4 && cidx + j < kred_total; ++j)
+          dw_ws[(int64_t)k * rsc + cidx + j] = etile[row * BN + ub + j];
+      } else {
+        const int64_t o = (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
+        *reinterpret_cast<float4*>(dw_ws + o) =
+            *reinterpret_cast<const float4*>(etile + row * BN + ub);
       }
     }
   }
