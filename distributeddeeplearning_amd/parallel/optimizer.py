@@ -107,6 +107,21 @@ class DistributedOptimizer:
             if b.handle is not None:
                 b.handle.wait()
                 b.handle = None
+        self._inspect_stall()
+
+    def _inspect_stall(self) -> None:
+        """Stall inspector (Horovod stall_inspector.cc role): warn when the
+        native engine reports collectives outstanding for too long."""
+        backend = api._backend
+        eng = getattr(backend, "engine", None)
+        if eng is not None and not self._warned_stall:
+            age = eng.oldest_age_seconds()
+            if age > STALL_WARN_SECONDS:
+                self._warned_stall = True
+                warnings.warn(
+                    f"collective outstanding for {age:.0f}s "
+                    f"({eng.outstanding()} in flight) — possible rank "
+                    "desync/deadlock")
 
     def step(self, closure=None):
         self._passes += 1
