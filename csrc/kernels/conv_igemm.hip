@@ -6,14 +6,14 @@
 //                                          w[k,r,s,c]   (terms where /sh,/sw
 //                                          divide exactly)
 // Both are the same GEMM pipeline over the linearized (r,s,c-tile) reduce
-// sequence: double-buffered swizzled LDS tiles with register staging split
-// issue-early/write-late (guide §6 G15 T14: next tile's global loads issue
-// before the MFMA section, the LDS write lands after it, one barrier per
-// tile), 128x128x64 block tile, 4 waves, v_mfma_f32_16x16x32_{bf16,f16}.
-// Padding is handled by per-row validity masks (zero-fill); dgrad consumes
-// the CRSK-transposed weight (ddl_transpose_krsc_crsk) so its B-tile loads
-// are contiguous.  MODE_FLAT flattens the reduce dim to R*S*C with scalar
-// gather for C % 8 != 0 (the 7x7 C=3 stem).
+// sequence: 128x128x64 block tile, 4 waves, v_mfma_f32_16x16x32_{bf16,f16},
+// operand tiles staged by global_load_lds (async HBM->LDS DMA, no VGPR
+// round-trip — guide §5 step 3; +67% over register staging on the plain
+// GEMM ladder).  glds writes lane-linear, so the bank swizzle lives on the
+// per-lane SOURCE address (rule 21): column unit cu -> cu ^ (row & 7),
+// giving <=2-way b128 read conflicts (guide G4).  Padding rows read a
+// device zero page (glds has no predication).  MODE_FLAT (C % 8 != 0, the
+// 7x7 C=3 stem) keeps register staging with per-element scalar gather.
 #include "conv_tile.h"
 #include "api.h"
 
@@ -21,11 +21,22 @@ namespace {
 
 enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2 };
 
-// DBUF=true: double-buffered LDS (64 KiB, 2 blocks/CU) with loads issued
-// under the MFMAs — best for deep reduce loops.  DBUF=false: single buffer
-// (32 KiB, ~4 blocks/CU) with issue-early loads and two barriers per tile —
-// the shallow-reduce 56x56/64-channel layers are LATENCY-bound (their whole
-// K-loop is 1-9 tiles of ~600 MFMA cycles), so occupancy beats pipelining.
+__device__ __align__(256) const unsigned char ddl_zero_page[256] = {0};
+
+// igemm-local swizzle: bits 4-6 only so the glds source XOR stays inside a
+// row's 128 B span.
+__device__ __forceinline__ int swz7_off(int row, int colbyte) {
+  return ((row << 7) + colbyte) ^ ((row & 7) << 4);
+}
+
+typedef __attribute__((address_space(1))) const unsigned int as1_uint;
+typedef __attribute__((address_space(3))) unsigned int as3_uint;
+
+__device__ __forceinline__ void glds16(const void* gsrc, void* lds_base) {
+  __builtin_amdgcn_global_load_lds((as1_uint*)gsrc, (as3_uint*)lds_base, 16,
+                                   0, 0);
+}
+
 template <typename T, int MODE, bool DBUF>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
@@ -42,15 +53,15 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   const int tile_n = (int)(bid % nt_n) * BN;
 
   const int lane = threadIdx.x & 63;
-  const int wid = threadIdx.x >> 6;
-  const int wr = (wid >> 1) * 64;
-  const int wc = (wid & 1) * 64;
-  const int trow = threadIdx.x >> 3;
-  const int tcol = threadIdx.x & 7;
+  const int wid = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+  const int wr = ((threadIdx.x >> 6) >> 1) * 64;
+  const int wc = ((threadIdx.x >> 6) & 1) * 64;
+  const int trow = threadIdx.x >> 3;   // this thread's staging row base
+  const int tcol = threadIdx.x & 7;    // 16B unit within the row
 
   f32x4_t acc[4][4] = {};
 
-  // per-thread A-row decode (4 staged rows)
+  // per-thread A-row decode (4 staged rows: trow + 32*i)
   int a_n[4], a_i1[4], a_i2[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
@@ -81,95 +92,110 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   int cur_rs = -1;
   const T* a_ptr[4];
   const T* b_ptr[4];
+  const T* zpad = reinterpret_cast<const T*>(ddl_zero_page);
 
-  // ---- per-tile register staging (issue-early half) ----------------------
-  uint4 ra[4], rb[4];
-  auto load_tile = [&](int t) {
-    const int rs = t / nct;
-    const int c0 = (t % nct) * BK;
+  auto update_rs_ptrs = [&](int rs) {
     const int r = rs / d.S, s = rs % d.S;
-    if (rs != cur_rs) {
-      cur_rs = rs;
+    cur_rs = rs;
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        a_ptr[i] = nullptr;
-        if (a_n[i] >= 0) {
-          if (MODE == FWD) {
-            const int hi = a_i1[i] * d.sh + r - d.ph;
-            const int wi = a_i2[i] * d.sw + s - d.pw;
-            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-              a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
-          } else if (MODE == DGRAD) {
-            const int hn = a_i1[i] + d.ph - r;
-            const int wn = a_i2[i] + d.pw - s;
-            if (hn >= 0 && wn >= 0 && hn % d.sh == 0 && wn % d.sw == 0) {
-              const int ho = hn / d.sh, wo = wn / d.sw;
-              if (ho < d.P && wo < d.Q)
-                a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
-            }
-          } else {
-            a_ptr[i] = Asrc;
+    for (int i = 0; i < 4; ++i) {
+      a_ptr[i] = nullptr;
+      if (a_n[i] >= 0) {
+        if (MODE == FWD) {
+          const int hi = a_i1[i] * d.sh + r - d.ph;
+          const int wi = a_i2[i] * d.sw + s - d.pw;
+          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+            a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
+        } else if (MODE == DGRAD) {
+          const int hn = a_i1[i] + d.ph - r;
+          const int wn = a_i2[i] + d.pw - s;
+          if (hn >= 0 && wn >= 0 && hn % d.sh == 0 && wn % d.sw == 0) {
+            const int ho = hn / d.sh, wo = wn / d.sw;
+            if (ho < d.P && wo < d.Q)
+              a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
           }
-        }
-      }
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row_n = tile_n + trow + 32 * i;
-        b_ptr[i] = nullptr;
-        if (row_n < Nc) {
-          if (MODE == FWD_FLAT)
-            b_ptr[i] = Bsrc + (int64_t)row_n * kred_total;
-          else
-            b_ptr[i] = Bsrc + ((int64_t)row_n * d.R + r) * d.S * b_ld +
-                       (int64_t)s * b_ld;
+        } else {
+          a_ptr[i] = Asrc;
         }
       }
     }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row_n = tile_n + trow + 32 * i;
+      b_ptr[i] = nullptr;
+      if (row_n < Nc) {
+        if (MODE == FWD_FLAT)
+          b_ptr[i] = Bsrc + (int64_t)row_n * kred_total;
+        else
+          b_ptr[i] = Bsrc + ((int64_t)row_n * d.R + r) * d.S * b_ld +
+                     (int64_t)s * b_ld;
+      }
+    }
+  };
+
+  // ---- glds staging (FWD/DGRAD): one async 16B DMA per thread per row ----
+  // Lane-linear LDS image; source column unit XOR'd with (row&7) so the
+  // b128 fragment reads see the swizzled layout (rule 21).
+  const int scu = tcol ^ (trow & 7);   // this thread's swizzled source unit
+  auto stage_glds = [&](int t, char* la, char* lb) {
+    const int rs = t / nct;
+    const int c0 = (t % nct) * BK;
+    if (rs != cur_rs) update_rs_ptrs(rs);
+    const int cel = c0 + scu * 8;
+    const bool cok = cel + 8 <= Cred;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const T* src = (a_ptr[i] != nullptr && cok) ? a_ptr[i] + cel : zpad;
+      glds16(src, la + wid * 1024 + i * 4096);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const T* src = (b_ptr[i] != nullptr && cok) ? b_ptr[i] + cel : zpad;
+      glds16(src, lb + wid * 1024 + i * 4096);
+    }
+  };
+
+  // ---- register staging (FWD_FLAT only: scalar gather) -------------------
+  uint4 ra[4], rb[4];
+  auto load_tile_flat = [&](int t) {
+    const int c0 = (t % nct) * BK;
+    if (cur_rs != 0) update_rs_ptrs(0);
     const int cel = c0 + tcol * 8;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       ra[i] = zero16();
-      if (MODE == FWD_FLAT) {
-        if (a_n[i] >= 0) {
-          uint16_t* e = reinterpret_cast<uint16_t*>(&ra[i]);
+      if (a_n[i] >= 0) {
+        uint16_t* e = reinterpret_cast<uint16_t*>(&ra[i]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int kred = cel + j;
-            if (kred < kred_total) {
-              const int rr = kred / (d.S * d.C);
-              const int rem = kred % (d.S * d.C);
-              const int ss = rem / d.C;
-              const int cc = rem % d.C;
-              const int hi = a_i1[i] * d.sh + rr - d.ph;
-              const int wi = a_i2[i] * d.sw + ss - d.pw;
-              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                e[j] = *reinterpret_cast<const uint16_t*>(
-                    Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc);
-            }
+        for (int j = 0; j < 8; ++j) {
+          const int kred = cel + j;
+          if (kred < kred_total) {
+            const int rr = kred / (d.S * d.C);
+            const int rem = kred % (d.S * d.C);
+            const int ss = rem / d.C;
+            const int cc = rem % d.C;
+            const int hi = a_i1[i] * d.sh + rr - d.ph;
+            const int wi = a_i2[i] * d.sw + ss - d.pw;
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+              e[j] = *reinterpret_cast<const uint16_t*>(
+                  Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc);
           }
         }
-      } else if (a_ptr[i] != nullptr && cel + 8 <= Cred) {
-        ra[i] = *reinterpret_cast<const uint4*>(a_ptr[i] + cel);
       }
     }
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       rb[i] = zero16();
       if (b_ptr[i] != nullptr) {
-        if (MODE == FWD_FLAT) {
-          uint16_t* e = reinterpret_cast<uint16_t*>(&rb[i]);
+        uint16_t* e = reinterpret_cast<uint16_t*>(&rb[i]);
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (cel + j < kred_total)
-              e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
-        } else if (cel + 8 <= Cred) {
-          rb[i] = *reinterpret_cast<const uint4*>(b_ptr[i] + cel);
-        }
+        for (int j = 0; j < 8; ++j)
+          if (cel + j < kred_total)
+            e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
       }
     }
   };
-
-  auto write_tile = [&](char* la, char* lb) {
+  auto write_tile_flat = [&](char* la, char* lb) {
 #pragma unroll
     for (int i = 0; i < 4; ++i)
       lds_write16(la, trow + 32 * i, tcol * 16, ra[i]);
@@ -186,12 +212,20 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       typename MfmaFrag<T>::type a[4], b[4];
 #pragma unroll
       for (int mr = 0; mr < 4; ++mr) {
-        uint4 v = lds_read16(la, wr + mr * 16 + frow, kk * 64 + fcolb);
+        const int row = wr + mr * 16 + frow;
+        const int off = (MODE == FWD_FLAT)
+                            ? swz_off(row, kk * 64 + fcolb)
+                            : swz7_off(row, kk * 64 + fcolb);
+        uint4 v = *reinterpret_cast<const uint4*>(la + off);
         a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
       }
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr) {
-        uint4 v = lds_read16(lb, wc + nr * 16 + frow, kk * 64 + fcolb);
+        const int row = wc + nr * 16 + frow;
+        const int off = (MODE == FWD_FLAT)
+                            ? swz_off(row, kk * 64 + fcolb)
+                            : swz7_off(row, kk * 64 + fcolb);
+        uint4 v = *reinterpret_cast<const uint4*>(lb + off);
         b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
       }
 #pragma unroll
@@ -202,38 +236,38 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   };
 
-  if (DBUF) {
-    // ---- pipelined main loop: one barrier per tile -----------------------
-    load_tile(0);
-    write_tile(lA[0], lB[0]);
+  if (MODE == FWD_FLAT) {
+    load_tile_flat(0);
+    for (int t = 0; t < total_tiles; ++t) {
+      write_tile_flat(lA[0], lB[0]);
+      __syncthreads();
+      if (t + 1 < total_tiles) load_tile_flat(t + 1);
+      mfma_tile(lA[0], lB[0]);
+      __syncthreads();
+    }
+  } else if (DBUF) {
+    // glds(t+1) into the other buffer runs under mfma(t); the barrier's
+    // implicit vmcnt(0) drain lands the DMA before its first read.
+    stage_glds(0, lA[0], lB[0]);
     __syncthreads();
     for (int t = 0; t < total_tiles; ++t) {
       const int cur = t & 1;
-      const bool more = (t + 1) < total_tiles;
-      if (more) load_tile(t + 1);               // global loads in flight...
-      char* la = cur ? lA[1] : lA[0];
-      char* lb = cur ? lB[1] : lB[0];
-      mfma_tile(la, lb);                        // ...hidden under the MFMAs
-      if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      if (t + 1 < total_tiles)
+        stage_glds(t + 1, cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
       __syncthreads();
     }
   } else {
-    // ---- single buffer, issue-early loads, two barriers per tile ---------
-    load_tile(0);
     for (int t = 0; t < total_tiles; ++t) {
-      write_tile(lA[0], lB[0]);
-      __syncthreads();
-      if (t + 1 < total_tiles) load_tile(t + 1);  // overlap with MFMAs
+      stage_glds(t, lA[0], lB[0]);
+      __syncthreads();  // drains the DMA (vmcnt(0) + barrier)
       mfma_tile(lA[0], lB[0]);
       __syncthreads();
     }
   }
 
-  // ---- epilogue through LDS: repack D fragments ([4 consecutive m] x
-  // [1 n] per lane) into a [128][128] element tile, then store coalesced
-  // 16B rows.  Direct fragment stores are 64 scalar 2B globals per lane —
-  // issue-bound (guide T21 diagnostic) and dominant for shallow layers.
-  __syncthreads();  // done with the operand tiles; reuse smem
+  // ---- epilogue through LDS: repack D fragments, store coalesced ---------
+  __syncthreads();
   T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
@@ -253,13 +287,12 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
   __syncthreads();
-  // 256 threads x 8 iterations of 16B: row-major sweep of the 128x128 tile
-  const int nvalid = Nc - tile_n;  // columns of this tile in range
+  const int nvalid = Nc - tile_n;
 #pragma unroll
   for (int it = 0; it < 8; ++it) {
-    const int unit = threadIdx.x + it * 256;   // 16B unit index (2048 total)
-    const int row = unit >> 4;                 // BN*2/16 = 16 units per row
-    const int ub = (unit & 15) * 8;            // first element in row
+    const int unit = threadIdx.x + it * 256;
+    const int row = unit >> 4;
+    const int ub = (unit & 15) * 8;
     const int64_t m = tile_m + row;
     if (m < M && ub < nvalid) {
       *reinterpret_cast<uint4*>(out + m * Nc + tile_n + ub) =
@@ -271,10 +304,6 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
 bool conv_use_dbuf(int cred, int rs_total) {
   const char* env = getenv("DDL_CONV_DBUF");
   if (env) return env[0] == '1';
-  // measured (bench ladder): the 32 KiB single-buffer variant's extra
-  // occupancy (4 blocks/CU vs 2) beats the double-buffer pipeline at every
-  // ResNet/VGG shape on gfx950 — these kernels are latency-bound, not
-  // bandwidth-bound (PMC: MfmaUtil 19%, VALUBusy 31%).
   (void)cred; (void)rs_total;
   return false;
 }
