@@ -68,3 +68,30 @@ __device__ __forceinline__ uint4 zero16() { return uint4{0, 0, 0, 0}; }
 struct ConvDims {
   int N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw;
 };
+
+// Granlund-Montgomery round-up magic division (valid for any uint32 x):
+// m = ceil(2^(32+s)/d), q = (x*m) >> (32+s).  Integer divides in per-tile
+// address decode cost ~12 v_rcp-based sequences per staged tile otherwise.
+struct FastDiv {
+  uint32_t mul;
+  int shift;  // total shift = 32 + shift
+};
+
+inline FastDiv make_fastdiv(uint32_t d) {
+  FastDiv f;
+  int s = 0;
+  while ((1ull << s) < d) ++s;
+  if ((d & (d - 1)) == 0) {  // power of two: ceil(2^(32+s)/d) == 2^32
+    f.mul = 0;               // sentinel: plain shift
+    f.shift = s;
+    return f;
+  }
+  f.mul = (uint32_t)(((((unsigned __int128)1) << (32 + s)) + d - 1) / d);
+  f.shift = s;
+  return f;
+}
+
+__device__ __forceinline__ uint32_t fd_div(uint32_t x, FastDiv f) {
+  if (f.mul == 0) return x >> f.shift;
+  return (uint32_t)(((uint64_t)x * f.mul) >> 32 >> f.shift);
+}

@@ -15,18 +15,19 @@ enum WMode { TILED = 0, FLAT = 1 };
 
 // SIMPLE = 1x1/stride-1/no-pad conv: the x operand is just row m of the
 // input (no im2col decode; the per-tile q/p/n divides were a hot cost).
-template <typename T, int MODE, bool SIMPLE>
+template <typename T, int MODE, bool SIMPLE, bool DBUF>
 __global__ __launch_bounds__(256) void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
     float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows,
-    int64_t slab_len) {
+    int64_t slab_len, FastDiv fdq, FastDiv fdpq) {
   // each m-chunk (blockIdx.z) owns a private fp32 slab: plain dense stores,
   // no atomics (a chunks-deep atomicAdd chain per dw element serializes at
   // L2 like the BN-stats case), and the final sum is deterministic.
   dw_ws += (int64_t)blockIdx.z * slab_len;
-  __shared__ __align__(16) char smem[4 * TILE_BYTES];  // A0 B0 A1 B1
-  char* const lA[2] = {smem, smem + 2 * TILE_BYTES};   // dy^T: [128 k][64 m]
-  char* const lB[2] = {smem + TILE_BYTES, smem + 3 * TILE_BYTES};  // x^T
+  __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
+  char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
+  char* const lB[2] = {smem + TILE_BYTES,
+                       DBUF ? smem + 3 * TILE_BYTES : smem + TILE_BYTES};
 
   const int kred_total = (MODE == FLAT) ? d.R * d.S * d.C : d.C;
   const int ntc = (kred_total + BN - 1) / BN;
@@ -72,9 +73,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
         if (m < m_end && c0 + 8 <= d.C)
           vb[mm] = *reinterpret_cast<const uint4*>(x + m * d.C + c0);
       } else if (m < m_end) {
-        const int q = (int)(m % d.Q);
-        const int p = (int)((m / d.Q) % d.P);
-        const int n = (int)(m / ((int64_t)d.Q * d.P));
+        const uint32_t mq = fd_div((uint32_t)m, fdq);
+        const int q = (int)((uint32_t)m - mq * d.Q);
+        const uint32_t n = fd_div((uint32_t)m, fdpq);
+        const int p = (int)(mq - n * d.P);
         if (MODE == TILED) {
           const int hi = p * d.sh + r - d.ph;
           const int wi = q * d.sw + s - d.pw;
@@ -151,55 +153,71 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   };
 
   if (ntiles > 0) {
-    load_tile(0);
-    write_tile(lA[0], lB[0]);
-    __syncthreads();
-    for (int64_t t = 0; t < ntiles; ++t) {
-      const bool more = (t + 1) < ntiles;
-      const int cur = (int)(t & 1);
-      if (more) load_tile(t + 1);
-      mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
-      if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+    if (DBUF) {
+      load_tile(0);
+      write_tile(lA[0], lB[0]);
       __syncthreads();
-    }
-  }
-
-  // ---- epilogue: repack through LDS (fp32 [128][128] = the whole 64 KiB
-  // smem), then coalesced 16B stores into this chunk's fp32 slab ----------
-  __syncthreads();
-  float* etile = reinterpret_cast<float*>(smem);
-  const int en = lane & 15;
-  const int em = (lane >> 4) * 4;
-  const int rsc = d.R * d.S * d.C;
-#pragma unroll
-  for (int mr = 0; mr < 4; ++mr) {
-#pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-#pragma unroll
-      for (int nr = 0; nr < 4; ++nr) {
-        etile[(wr + mr * 16 + em + reg) * BN + wc + nr * 16 + en] =
-            acc[mr][nr][reg];
+      for (int64_t t = 0; t < ntiles; ++t) {
+        const bool more = (t + 1) < ntiles;
+        const int cur = (int)(t & 1);
+        if (more) load_tile(t + 1);
+        mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
+        if (more) write_tile(cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+        __syncthreads();
+      }
+    } else {
+      load_tile(0);
+      for (int64_t t = 0; t < ntiles; ++t) {
+        write_tile(lA[0], lB[0]);
+        __syncthreads();
+        if (t + 1 < ntiles) load_tile(t + 1);
+        mfma_tile(lA[0], lB[0]);
+        __syncthreads();
       }
     }
   }
-  __syncthreads();
+
+  // ---- epilogue: repack through LDS (fp32; SBUF smem fits half the tile
+  // at a time), then coalesced 16B stores into this chunk's fp32 slab -----
+  const int rsc = d.R * d.S * d.C;
   const int nvalid = kred_total - tile_c;
+  float* etile = reinterpret_cast<float*>(smem);
+  const int en = lane & 15;
+  const int em = (lane >> 4) * 4;
+  const int halves = DBUF ? 1 : 2;   // SBUF: two passes of 64 k-rows
+  for (int h = 0; h < halves; ++h) {
+    __syncthreads();
+    const int rows_per = BM / halves;
 #pragma unroll
-  for (int it = 0; it < 16; ++it) {
-    const int unit = threadIdx.x + it * 256;   // 16B = 4 floats; 4096 units
-    const int row = unit >> 5;                 // 32 units per 128-f32 row
-    const int ub = (unit & 31) * 4;
-    const int k = tile_k + row;
-    if (k < d.K && ub < nvalid) {
-      const int cidx = tile_c + ub;
-      if (MODE == FLAT) {
-        // flat slab rows (R*S*C, e.g. 147) are not 16B-aligned: scalar
-        for (int j = 0; j < 4 && cidx + j < kred_total; ++j)
-          dw_ws[(int64_t)k * rsc + cidx + j] = etile[row * BN + ub + j];
-      } else {
-        const int64_t o = (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
-        *reinterpret_cast<float4*>(dw_ws + o) =
-            *reinterpret_cast<const float4*>(etile + row * BN + ub);
+    for (int mr = 0; mr < 4; ++mr) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int krow = wr + mr * 16 + em + reg;
+        if (krow < h * rows_per || krow >= (h + 1) * rows_per) continue;
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr) {
+          etile[(krow - h * rows_per) * BN + wc + nr * 16 + en] =
+              acc[mr][nr][reg];
+        }
+      }
+    }
+    __syncthreads();
+    const int nunits = rows_per * BN / 4 / 256;  // 16B units per thread
+    for (int it = 0; it < nunits; ++it) {
+      const int unit = threadIdx.x + it * 256;
+      const int row = unit >> 5;
+      const int ub = (unit & 31) * 4;
+      const int k = tile_k + h * rows_per + row;
+      if (k < d.K && ub < nvalid) {
+        const int cidx = tile_c + ub;
+        if (MODE == FLAT) {
+          for (int j = 0; j < 4 && cidx + j < kred_total; ++j)
+            dw_ws[(int64_t)k * rsc + cidx + j] = etile[row * BN + ub + j];
+        } else {
+          const int64_t o = (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
+          *reinterpret_cast<float4*>(dw_ws + o) =
+              *reinterpret_cast<const float4*>(etile + row * BN + ub);
+        }
       }
     }
   }
@@ -288,18 +306,26 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
             (unsigned)pl.chunks);
   const bool simple = (R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
                        pw == 0 && !flat);
-#define RUN(T, MODE, SIMP)                                                     \
-  conv_wgrad_k<T, MODE, SIMP><<<grid, 256, 0, s>>>(                            \
-      (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len)
+  const char* env = getenv("DDL_WGRAD_DBUF");
+  const bool dbuf = env ? env[0] == '1' : false;
+  FastDiv fdq = make_fastdiv((uint32_t)Q);
+  FastDiv fdpq = make_fastdiv((uint32_t)P * Q);
+#define RUN(T, MODE, SIMP, DB)                                                 \
+  conv_wgrad_k<T, MODE, SIMP, DB><<<grid, 256, 0, s>>>(                        \
+      (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
+      fdpq)
+#define SEL(T, MODE, SIMP)                                                     \
+  if (dbuf) RUN(T, MODE, SIMP, true); else RUN(T, MODE, SIMP, false)
   if (dtype == 1) {
-    if (flat) RUN(bf16, FLAT, false);
-    else if (simple) RUN(bf16, TILED, true);
-    else RUN(bf16, TILED, false);
+    if (flat) { SEL(bf16, FLAT, false); }
+    else if (simple) { SEL(bf16, TILED, true); }
+    else { SEL(bf16, TILED, false); }
   } else if (dtype == 2) {
-    if (flat) RUN(f16, FLAT, false);
-    else if (simple) RUN(f16, TILED, true);
-    else RUN(f16, TILED, false);
+    if (flat) { SEL(f16, FLAT, false); }
+    else if (simple) { SEL(f16, TILED, true); }
+    else { SEL(f16, TILED, false); }
   }
+#undef SEL
 #undef RUN
   int g = (int)cdiv(slab_len, 256);
 #define RED(T)                                                                 \
