@@ -79,13 +79,17 @@ struct FastDiv {
 
 inline FastDiv make_fastdiv(uint32_t d) {
   FastDiv f;
-  int s = 0;
-  while ((1ull << s) < d) ++s;
-  if ((d & (d - 1)) == 0) {  // power of two: ceil(2^(32+s)/d) == 2^32
-    f.mul = 0;               // sentinel: plain shift
+  if ((d & (d - 1)) == 0) {  // power of two: plain shift
+    int s = 0;
+    while ((1u << s) < d) ++s;
+    f.mul = 0;  // sentinel
     f.shift = s;
     return f;
   }
+  // s = floor(log2 d) keeps mul = ceil(2^(32+s)/d) < 2^32; round-up magic
+  // is then exact for all x < 2^31 (Granlund-Montgomery).
+  int s = 0;
+  while ((1ull << (s + 1)) <= d) ++s;
   f.mul = (uint32_t)(((((unsigned __int128)1) << (32 + s)) + d - 1) / d);
   f.shift = s;
   return f;
