@@ -246,12 +246,24 @@ def create_backend(comm: Optional[str] = None) -> CommBackend:
         )
 
     if comm == "auto":
-        comm = "native" if torch.cuda.is_available() else "torchdist"
+        comm = "native-or-torchdist" if torch.cuda.is_available() else "torchdist"
 
     if comm == "native":
         dev = _default_device_index()
         torch.cuda.set_device(dev)
         return NativeRcclBackend(dev)
+    elif comm == "native-or-torchdist":
+        dev = _default_device_index()
+        torch.cuda.set_device(dev)
+        try:
+            return NativeRcclBackend(dev)
+        except Exception as e:  # noqa: BLE001 — any init failure
+            import warnings
+            warnings.warn(
+                f"native RCCL engine init failed ({e!r}); falling back to "
+                "torch.distributed nccl (still RCCL, without the native "
+                "side-stream engine)")
+            return TorchDistBackend()
     elif comm == "torchdist":
         return TorchDistBackend()
     elif comm == "single":
