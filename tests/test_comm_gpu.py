@@ -68,3 +68,20 @@ def _rank2_worker_entry(rank, world):
     import socket
     # port comes via env set by dist_helpers; reuse MASTER_PORT
     return _rank2_worker(rank, world, int(os.environ["MASTER_PORT"]))
+
+
+def test_native_engine_timeline(tmp_path):
+    import distributeddeeplearning_amd._C_comm as C
+    import json
+    eng = C.CommEngine(0, 1, 0, C.get_unique_id())
+    path = str(tmp_path / "comm_tl.json")
+    eng.enable_timeline(path)
+    t = torch.ones(1 << 20, device="cuda")
+    h = eng.allreduce_async(t, True)
+    eng.wait_handle(h)
+    torch.cuda.synchronize()
+    eng.dump_timeline()
+    data = json.load(open(path))
+    names = [e["name"] for e in data["traceEvents"]]
+    assert any("allreduce" in n for n in names)
+    eng.shutdown()

@@ -68,3 +68,18 @@ def test_resnet_backward_cpu():
 def test_resnet152_structure():
     m = models.resnet152()
     assert len(m.layer3) == 36 and len(m.layer2) == 8
+
+
+def test_vgg16_backward_cpu():
+    import torch
+    from distributeddeeplearning_amd import models
+    from distributeddeeplearning_amd.ops import functional as DF
+    torch.manual_seed(0)
+    m = models.vgg16(num_classes=10)
+    m.train()
+    x = torch.randn(1, 3, 224, 224)
+    y = torch.randint(0, 10, (1,))
+    loss = DF.softmax_cross_entropy(m(x), y)
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert all(p.grad is not None for p in m.parameters())
