@@ -40,6 +40,7 @@ from .parallel.api import (  # noqa: F401
     broadcast_optimizer_state,
 )
 from .parallel.optimizer import DistributedOptimizer  # noqa: F401
+from .parallel.compression import Compression  # noqa: F401
 from . import models  # noqa: F401
 from . import ops  # noqa: F401
 from .config import TrainConfig, PRESETS  # noqa: F401

@@ -21,4 +21,5 @@ from .api import (  # noqa: F401
     broadcast_optimizer_state,
 )
 from .optimizer import DistributedOptimizer  # noqa: F401
+from .compression import Compression  # noqa: F401
 from .bucket import BucketManager, Bucket  # noqa: F401

@@ -22,6 +22,7 @@ import torch
 
 from . import api
 from .bucket import BucketManager
+from .compression import Compression
 
 STALL_WARN_SECONDS = float(os.environ.get("DDL_STALL_WARN_SECONDS", "60"))
 
@@ -41,12 +42,18 @@ class DistributedOptimizer:
         backward_passes_per_step: int = 1,
         bucket_bytes: Optional[int] = None,
         average: bool = True,
+        compression=Compression.none,
     ):
         self.optimizer = optimizer
         self.backward_passes_per_step = int(backward_passes_per_step)
         self._passes = 0
         self.average = average
+        self.compression = compression
         self._timeline = None
+        # debug collective-order check: hash the bucket
+        # launch sequence and compare across ranks every step
+        self._order_check = os.environ.get("DDL_DEBUG_ORDER", "0") == "1"
+        self._launch_order: list[int] = []
 
         if named_parameters is None:
             named_parameters = [
@@ -84,7 +91,21 @@ class DistributedOptimizer:
         if api.size() > 1:
             if self._timeline is not None:
                 self._timeline.mark(f"bucket{bucket.index}/launch")
-            bucket.handle = api.allreduce_async_(bucket.flat, average=self.average)
+            if self._order_check:
+                self._launch_order.append(bucket.index)
+            wire, ctx = self.compression.compress(bucket.flat)
+            h = api.allreduce_async_(wire, average=self.average)
+            if ctx is not None:
+                inner = h
+
+                class _DecompressHandle:
+                    def wait(self2):
+                        t = inner.wait()
+                        return self.compression.decompress(t, ctx)
+
+                bucket.handle = _DecompressHandle()
+            else:
+                bucket.handle = h
 
     # -- hvd surface --------------------------------------------------------
 
@@ -108,6 +129,16 @@ class DistributedOptimizer:
                 b.handle.wait()
                 b.handle = None
         self._inspect_stall()
+        if self._order_check and api.size() > 1 and self._launch_order:
+            import hashlib
+            dig = hashlib.sha1(str(self._launch_order).encode()).digest()[:8]
+            t = torch.tensor(list(dig), dtype=torch.float64)
+            mine = t.clone()
+            api.allreduce_(t, average=True)
+            if not torch.equal(t, mine):
+                raise RuntimeError(
+                    f"collective order diverged across ranks: {self._launch_order}")
+            self._launch_order.clear()
 
     def _inspect_stall(self) -> None:
         """Stall inspector (Horovod stall_inspector.cc role): warn when the

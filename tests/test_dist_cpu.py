@@ -160,3 +160,32 @@ def test_trainer_cpu_plumbing_world2():
     res = run_distributed(_w_trainer_smoke, world=2, timeout=300)
     assert torch.equal(res[0]["pall"], res[1]["pall"]), "ranks diverged"
     assert res[0]["loss"] == res[0]["loss"]  # finite
+
+
+def _w_compression(rank, world):
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import (Compression,
+                                                      DistributedOptimizer,
+                                                      api)
+    api.init("torchdist")
+    os.environ["DDL_DEBUG_ORDER"] = "1"  # collective-order checksum on
+    model = _make_mlp(seed=5)
+    api.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = DistributedOptimizer(
+        FusedSGD(model.parameters(), lr=0.05, momentum=0.9),
+        named_parameters=model.named_parameters(), bucket_bytes=4096,
+        compression=Compression.fp16)
+    for step in range(2):
+        g = torch.Generator().manual_seed(step)
+        x = torch.randn(4, 16, generator=g)
+        y = torch.randn(4, 4, generator=g)
+        opt.zero_grad()
+        torch.nn.functional.mse_loss(model(x), y).backward()
+        opt.step()
+    params = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    return {"params": params}
+
+
+def test_fp16_compression_keeps_ranks_synced():
+    res = run_distributed(_w_compression, world=2)
+    assert torch.equal(res[0]["params"], res[1]["params"])
