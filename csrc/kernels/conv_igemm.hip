@@ -137,9 +137,9 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   // Lane-linear LDS image; source column unit XOR'd with (row&7) so the
   // b128 fragment reads see the swizzled layout (rule 21).
   const int scu = tcol ^ (trow & 7);   // this thread's swizzled source unit
-  auto stage_glds = [&](int t, char* la, char* lb) {
-    const int rs = t / nct;
-    const int c0 = (t % nct) * BK;
+  // (rs, c0) are carried by the caller as counters — a per-tile t/nct
+  // integer division costs ~40 VALU cycles/tile and showed up in PMC.
+  auto stage_glds = [&](int rs, int c0, char* la, char* lb) {
     if (rs != cur_rs) update_rs_ptrs(rs);
     const int cel = c0 + scu * 8;
     const bool cok = cel + 8 <= Cred;
@@ -157,8 +157,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
 
   // ---- register staging (FWD_FLAT only: scalar gather) -------------------
   uint4 ra[4], rb[4];
-  auto load_tile_flat = [&](int t) {
-    const int c0 = (t % nct) * BK;
+  auto load_tile_flat = [&](int c0) {
     if (cur_rs != 0) update_rs_ptrs(0);
     const int cel = c0 + tcol * 8;
 #pragma unroll
@@ -236,30 +235,42 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   };
 
+  const int c_last = (nct - 1) * BK;
   if (MODE == FWD_FLAT) {
     load_tile_flat(0);
+    int c0n = 0;
     for (int t = 0; t < total_tiles; ++t) {
       write_tile_flat(lA[0], lB[0]);
       __syncthreads();
-      if (t + 1 < total_tiles) load_tile_flat(t + 1);
+      if (t + 1 < total_tiles) {
+        c0n += BK;
+        load_tile_flat(c0n);
+      }
       mfma_tile(lA[0], lB[0]);
       __syncthreads();
     }
   } else if (DBUF) {
     // glds(t+1) into the other buffer runs under mfma(t); the barrier's
     // implicit vmcnt(0) drain lands the DMA before its first read.
-    stage_glds(0, lA[0], lB[0]);
+    stage_glds(0, 0, lA[0], lB[0]);
     __syncthreads();
+    int rsn = 0, c0n = 0;  // counters for tile t+1
     for (int t = 0; t < total_tiles; ++t) {
       const int cur = t & 1;
-      if (t + 1 < total_tiles)
-        stage_glds(t + 1, cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      if (t + 1 < total_tiles) {
+        c0n += BK;
+        if (c0n >= kred_total) { c0n = 0; ++rsn; }
+        stage_glds(rsn, c0n, cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+      }
       mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
       __syncthreads();
     }
   } else {
+    int rsn = 0, c0n = 0;
     for (int t = 0; t < total_tiles; ++t) {
-      stage_glds(t, lA[0], lB[0]);
+      stage_glds(rsn, c0n, lA[0], lB[0]);
+      c0n += BK;
+      if (c0n >= kred_total) { c0n = 0; ++rsn; }
       __syncthreads();  // drains the DMA (vmcnt(0) + barrier)
       mfma_tile(lA[0], lB[0]);
       __syncthreads();

@@ -307,7 +307,7 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   const bool simple = (R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
                        pw == 0 && !flat);
   const char* env = getenv("DDL_WGRAD_DBUF");
-  const bool dbuf = env ? env[0] == '1' : false;
+  const bool dbuf = env ? env[0] == '1' : true;  // measured: dbuf wins here
   FastDiv fdq = make_fastdiv((uint32_t)Q);
   FastDiv fdpq = make_fastdiv((uint32_t)P * Q);
 #define RUN(T, MODE, SIMP, DB)                                                 \
