@@ -25,7 +25,7 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=512)
+    p.add_argument("--batch-size", type=int, default=1024)
     p.add_argument("--model", type=str, default="resnet50")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--image-size", type=int, default=224)

@@ -10,7 +10,7 @@ from __future__ import annotations
 import argparse
 import dataclasses
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional
 
 import torch

@@ -3,7 +3,6 @@ class.  One process per GPU; backward overlaps bucketed all-reduce; the only
 sync point is ``DistributedOptimizer.synchronize()`` inside ``step``."""
 from __future__ import annotations
 
-import math
 import os
 import time
 from typing import Iterator, Optional, Tuple

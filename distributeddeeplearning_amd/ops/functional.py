@@ -14,7 +14,7 @@ torchvision resnet50 inside ``HorovodPytorch/src/imagenet_pytorch_horovod.py``
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F

@@ -14,8 +14,7 @@ and biases) as required by the batch-8k config (BASELINE.json:9).
 """
 from __future__ import annotations
 
-import math
-from typing import Iterable, List, Optional
+from typing import Optional
 
 import torch
 
