@@ -250,20 +250,32 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       __syncthreads();
     }
   } else if (DBUF) {
-    // glds(t+1) into the other buffer runs under mfma(t); the barrier's
-    // implicit vmcnt(0) drain lands the DMA before its first read.
+    // Counted-vmcnt pipeline (guide §5 T3/T4 minimum form): raw s_barrier
+    // (no implicit vmcnt(0) drain) + `s_waitcnt vmcnt(8)` lets tile t+1's
+    // 8 glds DMAs stay in flight across the barrier while tile t computes.
+    // Per iteration: barrier #1 = everyone done READING buf[cur^1] (it is
+    // about to be overwritten); issue glds(t+1); vmcnt(8) = own glds(t)
+    // landed (FIFO; only t+1's 8 remain); barrier #2 = every wave's
+    // glds(t) writes visible; then MFMA on buf[cur].
     stage_glds(0, 0, lA[0], lB[0]);
-    __syncthreads();
     int rsn = 0, c0n = 0;  // counters for tile t+1
     for (int t = 0; t < total_tiles; ++t) {
       const int cur = t & 1;
-      if (t + 1 < total_tiles) {
+      const bool more = (t + 1) < total_tiles;
+      __builtin_amdgcn_s_barrier();
+      if (more) {
         c0n += BK;
         if (c0n >= kred_total) { c0n = 0; ++rsn; }
         stage_glds(rsn, c0n, cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
+      __builtin_amdgcn_s_barrier();
+      // raw s_barrier has no compiler memory-ordering semantics: pin the
+      // fragment ds_reads below it (guide §5.4 rule 18 analog)
+      __builtin_amdgcn_sched_barrier(0);
       mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
-      __syncthreads();
     }
   } else {
     int rsn = 0, c0n = 0;
