@@ -121,3 +121,15 @@ def test_launcher_spawns_and_fails_fast(tmp_path):
         capture_output=True, text=True, timeout=60)
     assert r.returncode == 3
     assert "rank 0 of 2" in r.stdout
+
+
+def test_tune_buckets_cpu_smoke(tmp_path):
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "tune_buckets.py"),
+         "--model", "resnet18", "--batch-size", "4", "--image-size", "32",
+         "--num-classes", "10", "--dtype", "fp32", "--steps", "2",
+         "--warmup", "1", "--sizes-mb", "1,4"],
+        capture_output=True, text=True, timeout=300,
+        env={**os.environ, "DDL_COMM": "single"})
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "best_bucket_mb" in r.stdout

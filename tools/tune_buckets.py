@@ -19,6 +19,8 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="resnet50")
     p.add_argument("--batch-size", type=int, default=256)
+    p.add_argument("--image-size", type=int, default=224)
+    p.add_argument("--num-classes", type=int, default=1000)
     p.add_argument("--dtype", default="bf16")
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
@@ -32,6 +34,8 @@ def main():
     results = {}
     for mb in [float(x) for x in args.sizes_mb.split(",")]:
         cfg = TrainConfig(model=args.model, batch_size=args.batch_size,
+                          image_size=args.image_size,
+                          num_classes=args.num_classes,
                           dtype=args.dtype, bucket_mb=mb, warmup_epochs=0)
         tr = Trainer(cfg, steps_per_epoch=10 ** 9)
         res = tr.bench(args.steps, args.warmup)
