@@ -104,8 +104,17 @@ class BucketManager:
                 self._param_bucket[id(p)] = (b, pi)
 
         self._names_rev = names_rev
+        # precomputed grad-view base pointers for the cheap aliasing guard
+        self._expected_ptr = {}
+        for b in self.buckets:
+            for p, off in zip(b.params, b.offsets):
+                self._expected_ptr[id(p)] = b.flat[off].data_ptr()
 
     # -- readiness protocol -------------------------------------------------
+
+    def ptr_of(self, p: torch.nn.Parameter) -> int:
+        """Expected storage pointer of p's grad view (aliasing guard)."""
+        return self._expected_ptr[id(p)]
 
     def mark_ready(self, p: torch.nn.Parameter) -> Optional[Bucket]:
         """Mark one param's grad complete; return the bucket if it is full."""

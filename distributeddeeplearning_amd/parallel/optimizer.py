@@ -78,6 +78,14 @@ class DistributedOptimizer:
     def _make_hook(self):
         def hook(p: torch.nn.Parameter):
             self._last_ready_time = time.monotonic()
+            g = p.grad
+            if g is None or g.data_ptr() != self.buckets.ptr_of(p):
+                # e.g. zero_grad(set_to_none=True) detached the flat view:
+                # reducing the stale bucket would be silently wrong.
+                raise RuntimeError(
+                    "param.grad no longer aliases its flat bucket view "
+                    "(was zero_grad(set_to_none=True) called on the model?). "
+                    "Use DistributedOptimizer.zero_grad() instead.")
             bucket = self.buckets.mark_ready(p)
             if bucket is not None and self._should_reduce():
                 self._launch(bucket)

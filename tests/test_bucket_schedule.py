@@ -69,3 +69,22 @@ def test_warmup_piecewise_lr():
     assert abs(s.get(31 * spe) - 0.08) < 1e-9
     assert abs(s.get(61 * spe) - 0.008) < 1e-9
     assert abs(s.get(81 * spe) - 0.0008) < 1e-9
+
+
+def test_set_to_none_guard():
+    import pytest
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("single")
+    m = _model()
+    opt = DistributedOptimizer(FusedSGD(m.parameters(), lr=0.1),
+                               named_parameters=m.named_parameters())
+    m(torch.randn(2, 64)).sum().backward()
+    opt.step()
+    m.zero_grad(set_to_none=True)  # detaches the flat views
+    with pytest.raises(RuntimeError, match="flat bucket view"):
+        m(torch.randn(2, 64)).sum().backward()
+    # recovery path: DistributedOptimizer.zero_grad rebinds
+    opt.zero_grad()
+    m(torch.randn(2, 64)).sum().backward()
+    opt.step()
