@@ -88,8 +88,6 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w,
                     int64_t ph, int64_t pw, bool relu) {
   check_cl(x, "x"); check_cl(w, "w");
   TORCH_CHECK(x.scalar_type() == w.scalar_type());
-  TORCH_CHECK(x.scalar_type() != at::kFloat,
-              "GPU conv supports bf16/fp16 (fp32 path not built; use bf16)");
   Conv g = conv_geom(x, w, sh, sw, ph, pw);
   auto y = empty_nhwc(g.N, g.K, g.P, g.Q, x.options());
   const void* bptr = nullptr;
@@ -98,7 +96,8 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w,
     bptr = bias->data_ptr();
   }
   int dt = dt_code(x);
-  if (g.C % 8 != 0) {
+  const int epu = dt == 0 ? 4 : 8;
+  if (g.C % epu != 0) {
     const int kred = g.R * g.S * g.C;
     const int ld = (kred + 7) & ~7;
     auto col = build_im2col(x, g, sh, sw, ph, pw, ld);
@@ -165,7 +164,8 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
   int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
   int K = w.size(0), R = w.size(2), S = w.size(3);
   int P = dy.size(2), Q = dy.size(3);
-  TORCH_CHECK(C % 8 == 0, "dgrad requires C % 8 == 0 (skip first conv)");
+  TORCH_CHECK(C % (dy.scalar_type() == at::kFloat ? 4 : 8) == 0,
+              "dgrad requires channel alignment (skip first conv)");
   // build CRSK-transposed weight for contiguous dgrad B-tiles
   auto w_t = at::empty({(int64_t)C * R * S * K}, w.options());
   ddl_transpose_krsc_crsk(w.data_ptr(), w_t.data_ptr(), K, R * S, C,
@@ -185,7 +185,7 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
   int P = dy.size(2), Q = dy.size(3);
   int dt = dt_code(x);
   auto dw = empty_nhwc(K, C, R, S, x.options());
-  if (C % 8 != 0) {
+  if (C % (dt == 0 ? 4 : 8) != 0) {
     const int kred = R * S * C;
     const int ld = (kred + 7) & ~7;
     Conv g; g.N = N; g.H = H; g.W = W; g.C = C; g.K = K; g.R = R; g.S = S;

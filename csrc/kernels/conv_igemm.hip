@@ -23,12 +23,6 @@ enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2 };
 
 __device__ __align__(256) const unsigned char ddl_zero_page[256] = {0};
 
-// igemm-local swizzle: bits 4-6 only so the glds source XOR stays inside a
-// row's 128 B span.
-__device__ __forceinline__ int swz7_off(int row, int colbyte) {
-  return ((row << 7) + colbyte) ^ ((row & 7) << 4);
-}
-
 typedef __attribute__((address_space(1))) const unsigned int as1_uint;
 typedef __attribute__((address_space(3))) unsigned int as3_uint;
 
@@ -84,9 +78,10 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   const int Cred = (MODE == DGRAD) ? d.K : d.C;
   const int64_t a_ld = (MODE == DGRAD) ? d.K : d.C;
   const int b_ld = (MODE == DGRAD) ? d.K : d.C;
+  constexpr int BKT = ConvElems<T>::BKT;  // reduce elements per LDS tile
   const int rs_total = (MODE == FWD_FLAT) ? 1 : d.R * d.S;
   const int kred_total = (MODE == FWD_FLAT) ? d.R * d.S * d.C : Cred;
-  const int nct = (kred_total + BK - 1) / BK;
+  const int nct = (kred_total + BKT - 1) / BKT;
   const int total_tiles = rs_total * nct;
 
   int cur_rs = -1;
@@ -136,13 +131,14 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   // ---- glds staging (FWD/DGRAD): one async 16B DMA per thread per row ----
   // Lane-linear LDS image; source column unit XOR'd with (row&7) so the
   // b128 fragment reads see the swizzled layout (rule 21).
+  constexpr int EPU = ConvElems<T>::PER_UNIT;  // elements per 16 B unit
   const int scu = tcol ^ (trow & 7);   // this thread's swizzled source unit
   // (rs, c0) are carried by the caller as counters — a per-tile t/nct
   // integer division costs ~40 VALU cycles/tile and showed up in PMC.
   auto stage_glds = [&](int rs, int c0, char* la, char* lb) {
     if (rs != cur_rs) update_rs_ptrs(rs);
-    const int cel = c0 + scu * 8;
-    const bool cok = cel + 8 <= Cred;
+    const int cel = c0 + scu * EPU;
+    const bool cok = cel + EPU <= Cred;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const T* src = (a_ptr[i] != nullptr && cok) ? a_ptr[i] + cel : zpad;
@@ -159,14 +155,14 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   uint4 ra[4], rb[4];
   auto load_tile_flat = [&](int c0) {
     if (cur_rs != 0) update_rs_ptrs(0);
-    const int cel = c0 + tcol * 8;
+    const int cel = c0 + tcol * EPU;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       ra[i] = zero16();
       if (a_n[i] >= 0) {
-        uint16_t* e = reinterpret_cast<uint16_t*>(&ra[i]);
+        T* e = reinterpret_cast<T*>(&ra[i]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int j = 0; j < EPU; ++j) {
           const int kred = cel + j;
           if (kred < kred_total) {
             const int rr = kred / (d.S * d.C);
@@ -176,8 +172,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
             const int hi = a_i1[i] * d.sh + rr - d.ph;
             const int wi = a_i2[i] * d.sw + ss - d.pw;
             if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-              e[j] = *reinterpret_cast<const uint16_t*>(
-                  Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc);
+              e[j] = Asrc[(((int64_t)a_n[i] * d.H + hi) * d.W + wi) * d.C + cc];
           }
         }
       }
@@ -186,11 +181,11 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     for (int i = 0; i < 4; ++i) {
       rb[i] = zero16();
       if (b_ptr[i] != nullptr) {
-        uint16_t* e = reinterpret_cast<uint16_t*>(&rb[i]);
+        T* e = reinterpret_cast<T*>(&rb[i]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
+        for (int j = 0; j < EPU; ++j)
           if (cel + j < kred_total)
-            e[j] = *reinterpret_cast<const uint16_t*>(b_ptr[i] + cel + j);
+            e[j] = b_ptr[i][cel + j];
       }
     }
   };
@@ -204,34 +199,53 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   };
 
   const int frow = lane & 15;
-  const int fcolb = (lane >> 4) * 16;
+  // 2-byte dtypes: 2 K=32 mfma steps per tile reading 16B fragments;
+  // fp32: 8 K=4 v_mfma_f32_16x16x4_f32 steps reading one scalar per lane
+  // (exact f32 at the vector rate — no TF32 on gfx950, guide §3).
+  auto lds_off = [&](int row, int colb) {
+    return (MODE == FWD_FLAT) ? swz_off(row, colb) : swz7_off(row, colb);
+  };
   auto mfma_tile = [&](const char* la, const char* lb) {
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      typename MfmaFrag<T>::type a[4], b[4];
+    for (int kk = 0; kk < ConvElems<T>::KSTEPS; ++kk) {
+      if constexpr (sizeof(T) == 2) {
+        const int fcolb = kk * 64 + (lane >> 4) * 16;
+        typename MfmaFrag<T>::type a[4], b[4];
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr) {
-        const int row = wr + mr * 16 + frow;
-        const int off = (MODE == FWD_FLAT)
-                            ? swz_off(row, kk * 64 + fcolb)
-                            : swz7_off(row, kk * 64 + fcolb);
-        uint4 v = *reinterpret_cast<const uint4*>(la + off);
-        a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-      }
+        for (int mr = 0; mr < 4; ++mr) {
+          uint4 v = *reinterpret_cast<const uint4*>(
+              la + lds_off(wr + mr * 16 + frow, fcolb));
+          a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
 #pragma unroll
-      for (int nr = 0; nr < 4; ++nr) {
-        const int row = wc + nr * 16 + frow;
-        const int off = (MODE == FWD_FLAT)
-                            ? swz_off(row, kk * 64 + fcolb)
-                            : swz7_off(row, kk * 64 + fcolb);
-        uint4 v = *reinterpret_cast<const uint4*>(lb + off);
-        b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-      }
+        for (int nr = 0; nr < 4; ++nr) {
+          uint4 v = *reinterpret_cast<const uint4*>(
+              lb + lds_off(wc + nr * 16 + frow, fcolb));
+          b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr)
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+      } else {
+        const int fcolb = kk * 16 + (lane >> 4) * 4;
+        float a[4], b[4];
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+          a[mr] = *reinterpret_cast<const float*>(
+              la + lds_off(wr + mr * 16 + frow, fcolb));
 #pragma unroll
         for (int nr = 0; nr < 4; ++nr)
-          acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+          b[nr] = *reinterpret_cast<const float*>(
+              lb + lds_off(wc + nr * 16 + frow, fcolb));
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a[mr], b[nr], acc[mr][nr], 0, 0, 0);
+      }
     }
   };
 
@@ -243,7 +257,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       write_tile_flat(lA[0], lB[0]);
       __syncthreads();
       if (t + 1 < total_tiles) {
-        c0n += BK;
+        c0n += BKT;
         load_tile_flat(c0n);
       }
       mfma_tile(lA[0], lB[0]);
@@ -264,7 +278,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       const bool more = (t + 1) < total_tiles;
       __builtin_amdgcn_s_barrier();
       if (more) {
-        c0n += BK;
+        c0n += BKT;
         if (c0n >= kred_total) { c0n = 0; ++rsn; }
         stage_glds(rsn, c0n, cur ? lA[0] : lA[1], cur ? lB[0] : lB[1]);
         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
@@ -281,7 +295,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     int rsn = 0, c0n = 0;
     for (int t = 0; t < total_tiles; ++t) {
       stage_glds(rsn, c0n, lA[0], lB[0]);
-      c0n += BK;
+      c0n += BKT;
       if (c0n >= kred_total) { c0n = 0; ++rsn; }
       __syncthreads();  // drains the DMA (vmcnt(0) + barrier)
       mfma_tile(lA[0], lB[0]);
@@ -289,11 +303,34 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
 
-  // ---- epilogue through LDS: repack D fragments, store coalesced ---------
-  __syncthreads();
-  T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
+  // ---- epilogue -----------------------------------------------------------
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
+  if constexpr (sizeof(T) == 4) {
+    // fp32: direct fragment stores (the repack tile would need 64 KiB)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      const int n = tile_n + wc + nr * 16 + en;
+      if (n >= Nc) continue;
+      const float bv = (MODE != DGRAD && bias != nullptr)
+                           ? to_f32(bias[n]) : 0.f;
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr) {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int64_t m = tile_m + wr + mr * 16 + em + reg;
+          if (m >= M) continue;
+          float f = acc[mr][nr][reg] + bv;
+          if (relu) f = fmaxf(f, 0.f);
+          out[m * Nc + n] = f;
+        }
+      }
+    }
+    return;
+  }
+  // 2-byte dtypes: repack through LDS, store coalesced 16 B rows
+  __syncthreads();
+  T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
 #pragma unroll
   for (int nr = 0; nr < 4; ++nr) {
     const int n_loc = wc + nr * 16 + en;
@@ -343,14 +380,15 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
   const int64_t M = (int64_t)N * P * Q;
   const int Nc = K;
   const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
-  const bool flat = (C % 8) != 0;
+  const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
   const bool dbuf = conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
 #define RUN(T, MODE, DB)                                                       \
   conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
       (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu)
 #define SEL(T, MODE)                                                           \
   if (dbuf) RUN(T, MODE, true); else RUN(T, MODE, false)
-  if (dtype == 1) { if (flat) { SEL(bf16, FWD_FLAT); } else { SEL(bf16, FWD); } }
+  if (dtype == 0) { if (flat) { SEL(float, FWD_FLAT); } else { SEL(float, FWD); } }
+  else if (dtype == 1) { if (flat) { SEL(bf16, FWD_FLAT); } else { SEL(bf16, FWD); } }
   else if (dtype == 2) { if (flat) { SEL(f16, FWD_FLAT); } else { SEL(f16, FWD); } }
 #undef SEL
 #undef RUN
@@ -367,7 +405,8 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
       (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0)
-  if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
+  if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
+  else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
   else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
 #undef RUN
 }

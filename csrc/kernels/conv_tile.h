@@ -21,6 +21,26 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 template <typename T> struct MfmaFrag;
 template <> struct MfmaFrag<bf16> { using type = bf16x8_t; };
 template <> struct MfmaFrag<f16> { using type = f16x8_t; };
+template <> struct MfmaFrag<float> { using type = float; };
+
+// Elements per 16-byte staging unit / per 128-byte LDS row.  The byte
+// geometry of every tile is dtype-invariant (rows are 128 B), so the
+// swizzle, glds staging and LDS footprint are shared: bf16/f16 rows hold
+// 64 elements (BK=64 reduce per tile, mfma K=32 x2), fp32 rows hold 32
+// (BK=32, v_mfma_f32_16x16x4_f32 x8 — exact f32 at the vector rate;
+// gfx950 has no TF32 shortcut, guide §3).
+template <typename T> struct ConvElems {
+  static constexpr int PER_UNIT = 16 / (int)sizeof(T);   // elements per 16 B
+  static constexpr int BKT = 128 / (int)sizeof(T);       // reduce per tile
+  static constexpr int KSTEPS = sizeof(T) == 4 ? 8 : 2;  // mfma issues/tile
+};
+
+// igemm-local swizzle (bits 4-6 only, so the glds source XOR stays inside
+// a row's 128 B span — rule 21); the wgrad register-staged kernel uses the
+// wider swz_off above.
+__device__ __forceinline__ int swz7_off(int row, int colbyte) {
+  return ((row << 7) + colbyte) ^ ((row & 7) << 4);
+}
 
 template <typename T>
 __device__ __forceinline__ f32x4_t mfma_16x16x32(typename MfmaFrag<T>::type a,

@@ -40,37 +40,42 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   const int wid = threadIdx.x >> 6;
   const int wr = (wid >> 1) * 64;
   const int wc = (wid & 1) * 64;
-  // staging assignment (in-register 8x4 transpose, b64 LDS writes): each
-  // thread owns an 8-wide k/c group x 4 consecutive m: 4 x 16B loads ->
-  // repack -> 8 x 8B transposed writes.  16 k-groups x 16 m-groups = 256.
-  const int su = threadIdx.x & 15;        // k/c group (0..15) -> 8 rows
-  const int sm4 = (threadIdx.x >> 4) * 4; // first of 4 owned m columns
+  // staging assignment (in-register transpose): each thread owns a
+  // 16B-wide k/c group x 4 consecutive m.  2-byte dtypes: 16 groups of 8
+  // rows x 16 m-quads -> 8 b64 transposed writes; fp32: 32 groups of 4
+  // rows x 8 m-quads -> 4 b128 transposed writes.
+  constexpr int EPU = ConvElems<T>::PER_UNIT;   // k rows per 16 B load
+  constexpr int NSU = 256 / EPU / 2;            // staging groups (16 / 32)
+  const int su = threadIdx.x & (NSU - 1);       // k/c group
+  const int sm4 = (threadIdx.x / NSU) * 4;      // first of 4 owned m columns
 
   f32x4_t acc[4][4] = {};
 
+  constexpr int BKT = ConvElems<T>::BKT / 2;  // m columns per LDS tile
   const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
   const int64_t m_end = min(M, m_begin + chunk_rows);
-  const int64_t ntiles = (m_end > m_begin) ? (m_end - m_begin + BK - 1) / BK : 0;
+  const int64_t ntiles =
+      (m_end > m_begin) ? (m_end - m_begin + BKT - 1) / BKT : 0;
 
   uint4 va[4], vb[4];
   auto load_tile = [&](int64_t ti) {
-    const int64_t m0 = m_begin + ti * BK;
-    const int k0g = tile_k + su * 8;
+    const int64_t m0 = m_begin + ti * BKT;
+    const int k0g = tile_k + su * EPU;
 #pragma unroll
     for (int mm = 0; mm < 4; ++mm) {
       const int64_t m = m0 + sm4 + mm;
       va[mm] = zero16();
-      if (m < m_end && k0g + 8 <= d.K)
+      if (m < m_end && k0g + EPU <= d.K)
         va[mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
     }
 #pragma unroll
     for (int mm = 0; mm < 4; ++mm) {
       const int64_t m = m0 + sm4 + mm;
       vb[mm] = zero16();
-      uint16_t* e = reinterpret_cast<uint16_t*>(&vb[mm]);
+      T* e = reinterpret_cast<T*>(&vb[mm]);
       if (SIMPLE) {
-        const int c0 = tile_c + su * 8;
-        if (m < m_end && c0 + 8 <= d.C)
+        const int c0 = tile_c + su * EPU;
+        if (m < m_end && c0 + EPU <= d.C)
           vb[mm] = *reinterpret_cast<const uint4*>(x + m * d.C + c0);
       } else if (m < m_end) {
         const uint32_t mq = fd_div((uint32_t)m, fdq);
@@ -80,14 +85,14 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
         if (MODE == TILED) {
           const int hi = p * d.sh + r - d.ph;
           const int wi = q * d.sw + s - d.pw;
-          const int c0 = tile_c + su * 8;
-          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + 8 <= d.C)
+          const int c0 = tile_c + su * EPU;
+          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + EPU <= d.C)
             vb[mm] = *reinterpret_cast<const uint4*>(
                 x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
         } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int kred = tile_c + su * 8 + j;
+          for (int j = 0; j < EPU; ++j) {
+            const int kred = tile_c + su * EPU + j;
             if (kred < kred_total) {
               const int rr = kred / (d.S * d.C);
               const int rem = kred % (d.S * d.C);
@@ -96,8 +101,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
               const int hi = p * d.sh + rr - d.ph;
               const int wi = q * d.sw + ss - d.pw;
               if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                e[j] = *reinterpret_cast<const uint16_t*>(
-                    x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc);
+                e[j] = x[(((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc];
             }
           }
         }
@@ -105,50 +109,75 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
     }
   };
 
+  auto transpose_write = [&](char* lds, const uint4 (&v)[4]) {
+    if constexpr (sizeof(T) == 2) {
+      const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
+      const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
+      const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
+      const uint16_t* e3 = reinterpret_cast<const uint16_t*>(&v[3]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
+                      ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
+        *reinterpret_cast<uint64_t*>(lds + swz_off(su * 8 + j, sm4 * 2)) = pk;
+      }
+    } else {
+      const float* e0 = reinterpret_cast<const float*>(&v[0]);
+      const float* e1 = reinterpret_cast<const float*>(&v[1]);
+      const float* e2 = reinterpret_cast<const float*>(&v[2]);
+      const float* e3 = reinterpret_cast<const float*>(&v[3]);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float4 pk{e0[j], e1[j], e2[j], e3[j]};
+        *reinterpret_cast<float4*>(lds + swz_off(su * 4 + j, sm4 * 4)) = pk;
+      }
+    }
+  };
   auto write_tile = [&](char* la, char* lb) {
-    const uint16_t* a0 = reinterpret_cast<const uint16_t*>(&va[0]);
-    const uint16_t* a1 = reinterpret_cast<const uint16_t*>(&va[1]);
-    const uint16_t* a2 = reinterpret_cast<const uint16_t*>(&va[2]);
-    const uint16_t* a3 = reinterpret_cast<const uint16_t*>(&va[3]);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      uint64_t pk = (uint64_t)a0[j] | ((uint64_t)a1[j] << 16) |
-                    ((uint64_t)a2[j] << 32) | ((uint64_t)a3[j] << 48);
-      *reinterpret_cast<uint64_t*>(la + swz_off(su * 8 + j, sm4 * 2)) = pk;
-    }
-    const uint16_t* b0 = reinterpret_cast<const uint16_t*>(&vb[0]);
-    const uint16_t* b1 = reinterpret_cast<const uint16_t*>(&vb[1]);
-    const uint16_t* b2 = reinterpret_cast<const uint16_t*>(&vb[2]);
-    const uint16_t* b3 = reinterpret_cast<const uint16_t*>(&vb[3]);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      uint64_t pk = (uint64_t)b0[j] | ((uint64_t)b1[j] << 16) |
-                    ((uint64_t)b2[j] << 32) | ((uint64_t)b3[j] << 48);
-      *reinterpret_cast<uint64_t*>(lb + swz_off(su * 8 + j, sm4 * 2)) = pk;
-    }
+    transpose_write(la, va);
+    transpose_write(lb, vb);
   };
 
   const int frow = lane & 15;
-  const int fcolb = (lane >> 4) * 16;
   auto mfma_tile = [&](const char* la, const char* lb) {
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      typename MfmaFrag<T>::type a[4], b[4];
+    for (int kk = 0; kk < ConvElems<T>::KSTEPS; ++kk) {
+      if constexpr (sizeof(T) == 2) {
+        const int fcolb = kk * 64 + (lane >> 4) * 16;
+        typename MfmaFrag<T>::type a[4], b[4];
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr) {
-        uint4 v = lds_read16(la, wr + mr * 16 + frow, kk * 64 + fcolb);
-        a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-      }
+        for (int mr = 0; mr < 4; ++mr) {
+          uint4 v = lds_read16(la, wr + mr * 16 + frow, fcolb);
+          a[mr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
 #pragma unroll
-      for (int nr = 0; nr < 4; ++nr) {
-        uint4 v = lds_read16(lb, wc + nr * 16 + frow, kk * 64 + fcolb);
-        b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
-      }
+        for (int nr = 0; nr < 4; ++nr) {
+          uint4 v = lds_read16(lb, wc + nr * 16 + frow, fcolb);
+          b[nr] = *reinterpret_cast<typename MfmaFrag<T>::type*>(&v);
+        }
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr)
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+      } else {
+        const int fcolb = kk * 16 + (lane >> 4) * 4;
+        float a[4], b[4];
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+          a[mr] = *reinterpret_cast<const float*>(
+              la + swz_off(wr + mr * 16 + frow, fcolb));
 #pragma unroll
         for (int nr = 0; nr < 4; ++nr)
-          acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+          b[nr] = *reinterpret_cast<const float*>(
+              lb + swz_off(wc + nr * 16 + frow, fcolb));
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                a[mr], b[nr], acc[mr][nr], 0, 0, 0);
+      }
     }
   };
 
@@ -299,7 +328,7 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
                     hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int64_t M = (int64_t)N * P * Q;
-  const bool flat = (C % 8) != 0;
+  const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
   WgradPlan pl = wgrad_plan(N, C, K, R, S, P, Q);
   const int64_t slab_len = (int64_t)K * R * S * C;
   dim3 grid((unsigned)(pl.ntk * pl.ntc), (unsigned)pl.rs_total,
@@ -316,7 +345,11 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
       fdpq)
 #define SEL(T, MODE, SIMP)                                                     \
   if (dbuf) RUN(T, MODE, SIMP, true); else RUN(T, MODE, SIMP, false)
-  if (dtype == 1) {
+  if (dtype == 0) {
+    if (flat) { SEL(float, FLAT, false); }
+    else if (simple) { SEL(float, TILED, true); }
+    else { SEL(float, TILED, false); }
+  } else if (dtype == 1) {
     if (flat) { SEL(bf16, FLAT, false); }
     else if (simple) { SEL(bf16, TILED, true); }
     else { SEL(bf16, TILED, false); }
@@ -331,7 +364,8 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
 #define RED(T)                                                                 \
   wgrad_reduce_cast_k<T><<<g, 256, 0, s>>>(dw_ws, (T*)dw, slab_len,            \
                                            (int)pl.chunks)
-  if (dtype == 1) RED(bf16);
+  if (dtype == 0) RED(float);
+  else if (dtype == 1) RED(bf16);
   else if (dtype == 2) RED(f16);
 #undef RED
 }

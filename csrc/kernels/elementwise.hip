@@ -157,28 +157,29 @@ __global__ void im2col_k(const T* __restrict__ x, T* __restrict__ out,
                          int C, int P, int Q, int sh, int sw, int ph, int pw,
                          int kred_total, int ldout) {
   const int64_t M = (int64_t)N * P * Q;
-  const int64_t total = M * (ldout / 8);
+  constexpr int EPU0 = 16 / (int)sizeof(T);
+  const int64_t total = M * (ldout / EPU0);
   for (int64_t i = global_tid(); i < total; i += grid_stride()) {
-    const int ug = (int)(i % (ldout / 8));
-    const int64_t m = i / (ldout / 8);
+    const int ug = (int)(i % (ldout / EPU0));
+    const int64_t m = i / (ldout / EPU0);
     const int q = (int)(m % Q);
     const int p = (int)((m / Q) % P);
     const int n = (int)(m / ((int64_t)Q * P));
     const int h0 = p * sh - ph, w0 = q * sw - pw;
-    uint16_t e[8] = {0};
+    constexpr int EPU = 16 / (int)sizeof(T);
+    T e[EPU] = {};
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int kred = ug * 8 + j;
+    for (int j = 0; j < EPU; ++j) {
+      const int kred = ug * EPU + j;
       if (kred < kred_total) {
         const int t = tbl[kred];
         const int hi = h0 + (t >> 20);
         const int wi = w0 + ((t >> 10) & 1023);
         if (hi >= 0 && hi < H && wi >= 0 && wi < W)
-          e[j] = *reinterpret_cast<const uint16_t*>(
-              x + (((int64_t)n * H + hi) * W + wi) * C + (t & 1023));
+          e[j] = x[(((int64_t)n * H + hi) * W + wi) * C + (t & 1023)];
       }
     }
-    *reinterpret_cast<uint4*>(out + m * ldout + ug * 8) =
+    *reinterpret_cast<uint4*>(out + m * ldout + ug * EPU) =
         *reinterpret_cast<const uint4*>(e);
   }
 }
@@ -280,7 +281,7 @@ void ddl_im2col(const void* x, void* out, const int* tbl, int N, int H,
 #define RUN(T)                                                                 \
   im2col_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, tbl, N, H, W, C, P,     \
                                 Q, sh, sw, ph, pw, kred_total, ldout)
-  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }
 
@@ -288,7 +289,7 @@ void ddl_pad_cols(const void* w, void* out, int K, int kred, int ldout,
                   int dtype, hipStream_t s) {
   int g = (int)std::min<int64_t>(cdiv((int64_t)K * ldout, 256), 1024);
 #define RUN(T) pad_cols_k<T><<<g, 256, 0, s>>>((const T*)w, (T*)out, K, kred, ldout)
-  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }
 
@@ -296,7 +297,7 @@ void ddl_narrow_cols(const void* w, void* out, int K, int kred, int ldout,
                      int dtype, hipStream_t s) {
   int g = (int)std::min<int64_t>(cdiv((int64_t)K * kred, 256), 1024);
 #define RUN(T) narrow_cols_k<T><<<g, 256, 0, s>>>((const T*)w, (T*)out, K, kred, ldout)
-  switch (dtype) { case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }
 

@@ -42,7 +42,8 @@ class _Conv2dFn(torch.autograd.Function):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         weight = to_channels_last(weight)
-        ctx.small_c = x.shape[1] % 8 != 0
+        epu = 4 if x.dtype == torch.float32 else 8
+        ctx.small_c = x.shape[1] % epu != 0
         if ctx.small_c:
             # stem path: keep the im2col buffer for wgrad reuse
             y, col = ext.conv_fwd_smallc(

@@ -119,3 +119,45 @@ def test_inference_matches_train_path_stats():
         x.float().cpu(), bn.running_mean.cpu(), bn.running_var.cpu(),
         bn.weight.float().cpu(), bn.bias.float().cpu(), False, 0.1, 1e-5))
     assert relerr(y, ref) < 0.02
+
+
+@pytest.mark.parametrize("n,c,h,w,k,r,s,p", [
+    (2, 64, 28, 28, 64, 1, 1, 0),
+    (2, 64, 28, 28, 128, 3, 2, 1),
+    (2, 3, 64, 64, 64, 7, 2, 3),   # fp32 stem via im2col
+])
+def test_conv_fp32_exact_mfma(n, c, h, w, k, r, s, p):
+    """fp32 conv on v_mfma_f32_16x16x4_f32: exact f32 fmaf-chain numerics
+    (tight tolerance vs torch fp32 — only summation order differs)."""
+    torch.manual_seed(11)
+    x = torch.randn(n, c, h, w).requires_grad_(True)
+    wt = (torch.randn(k, c, r, r) * 0.05).requires_grad_(True)
+    yr = F.conv2d(x, wt, None, s, p)
+    dy = torch.randn_like(yr)
+    yr.backward(dy)
+    from distributeddeeplearning_amd.ops import functional as DF
+    xg = cl(x.detach().to(DEV)).requires_grad_(True)
+    wg = cl(wt.detach().to(DEV)).requires_grad_(True)
+    y = DF.conv2d(xg, wg, None, s, p)
+    assert relerr(y, yr) < 2e-5, "fp32 fwd"
+    y.backward(cl(dy.to(DEV)))
+    if c % 4 == 0:
+        assert relerr(xg.grad, x.grad) < 2e-5, "fp32 dgrad"
+    assert relerr(wg.grad, wt.grad) < 5e-5, "fp32 wgrad"
+
+
+def test_resnet18_fp32_gpu_step():
+    from distributeddeeplearning_amd.config import TrainConfig
+    from distributeddeeplearning_amd.engine.trainer import Trainer
+    cfg = TrainConfig(model="resnet18", image_size=64, num_classes=100,
+                      batch_size=8, dtype="fp32", warmup_epochs=0,
+                      comm="single")
+    tr = Trainer(cfg, steps_per_epoch=10)
+    it = tr.data_iter()
+    losses = []
+    for _ in range(6):
+        x, y = next(it)
+        losses.append(float(tr.train_step(x, y).detach().float()))
+    torch.cuda.synchronize()
+    assert all(l == l for l in losses)
+    assert min(losses[-2:]) < losses[0]
