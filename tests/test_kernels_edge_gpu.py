@@ -136,7 +136,9 @@ def test_conv_fp32_exact_mfma(n, c, h, w, k, r, s, p):
     dy = torch.randn_like(yr)
     yr.backward(dy)
     from distributeddeeplearning_amd.ops import functional as DF
-    xg = cl(x.detach().to(DEV)).requires_grad_(True)
+    xg = cl(x.detach().to(DEV))
+    if c % 4 == 0:
+        xg.requires_grad_(True)  # stem (C=3) has no input grad by design
     wg = cl(wt.detach().to(DEV)).requires_grad_(True)
     y = DF.conv2d(xg, wg, None, s, p)
     assert relerr(y, yr) < 2e-5, "fp32 fwd"
