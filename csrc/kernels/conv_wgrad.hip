@@ -51,7 +51,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
 
   f32x4_t acc[4][4] = {};
 
-  constexpr int BKT = ConvElems<T>::BKT / 2;  // m columns per LDS tile
+  constexpr int BKT = ConvElems<T>::BKT;  // m columns per LDS tile (64 / 32)
   const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
   const int64_t m_end = min(M, m_begin + chunk_rows);
   const int64_t ntiles =
