@@ -39,8 +39,7 @@ class FusedSGD(torch.optim.Optimizer):
                         lars_eps=lars_eps)
         super().__init__(params, defaults)
         self.grad_scale = grad_scale  # multiply grads by 1/scale during update
-        self._table = None  # cached GPU chunk table
-        self._table_sig = None
+        self._tables = {}  # per-param-group cached GPU chunk tables
 
     # ---- state ------------------------------------------------------------
 
@@ -101,32 +100,35 @@ class FusedSGD(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
-        for group in self.param_groups:
+        for gi, group in enumerate(self.param_groups):
             params = [p for p in group["params"] if p.grad is not None]
             if not params:
                 continue
             if params[0].is_cuda:
-                self._step_gpu(group)
+                self._step_gpu(gi, group)
             else:
                 self._step_cpu(group)
         return loss
 
-    def _step_gpu(self, group):
+    def _step_gpu(self, gi, group):
         ext = load_kernels_ext()
         sig = self._sig(group)
-        if self._table is None or self._table_sig != sig:
-            self._table, self._ntensors = self._build_table(group)
-            self._table_sig = sig
-            self._norms = torch.zeros(2, self._ntensors, dtype=torch.float32,
-                                      device=self._table.device)
+        cached = self._tables.get(gi)
+        if cached is None or cached[0] != sig:
+            table, ntensors = self._build_table(group)
+            norms = torch.zeros(2, ntensors, dtype=torch.float32,
+                                device=table.device)
+            cached = (sig, table, norms)
+            self._tables[gi] = cached
+        _, table, norms = cached
         dt = group["params"][0].dtype
         if group["lars"]:
-            self._norms.zero_()
-            ext.lars_norms(self._table, self._norms, _dtype_code(dt),
+            norms.zero_()
+            ext.lars_norms(table, norms, _dtype_code(dt),
                            float(group["weight_decay"]),
                            float(1.0 / self.grad_scale))
         ext.fused_sgd(
-            self._table, self._norms if group["lars"] else None,
+            table, norms if group["lars"] else None,
             _dtype_code(dt),
             float(group["lr"]), float(group["momentum"]),
             float(group["weight_decay"]), bool(group["nesterov"]),
@@ -166,8 +168,7 @@ class FusedSGD(torch.optim.Optimizer):
         return sd
 
     def invalidate_table(self):
-        self._table = None
-        self._table_sig = None
+        self._tables = {}
 
 
 def _dtype_code(dt: torch.dtype) -> int:
