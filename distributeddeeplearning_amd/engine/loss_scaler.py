@@ -8,8 +8,6 @@ from __future__ import annotations
 
 import torch
 
-from ..ops import functional as DF
-
 
 class DynamicLossScaler:
     def __init__(self, init_scale: float = 2.0 ** 16, growth_factor: float = 2.0,

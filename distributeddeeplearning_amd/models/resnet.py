@@ -10,9 +10,8 @@ GPU).  Shape table: SURVEY.md Appendix A.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Type, Union
+from typing import List, Type, Union
 
-import torch
 import torch.nn as nn
 
 from ..ops.modules import (

@@ -7,7 +7,6 @@ Physical layout on GPU is NHWC (channels_last); logical shapes stay NCHW.
 """
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -14,8 +14,6 @@ and biases) as required by the batch-8k config (BASELINE.json:9).
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from ._ext import load_kernels_ext

@@ -18,7 +18,7 @@ doesn't serialise after backward ends.  Tune with tools/tune_buckets.py.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 import torch

@@ -9,7 +9,6 @@ output in chrome://tracing or Perfetto.  Enable with ``DDL_TIMELINE=path`` or
 from __future__ import annotations
 
 import json
-import os
 import threading
 import time
 
