@@ -260,18 +260,23 @@ at::Tensor bn_relu_infer(at::Tensor x, at::Tensor weight, at::Tensor bias,
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
-    at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor mean,
-    at::Tensor rstd, at::Tensor y, bool relu, bool has_residual) {
+    at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
+    at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
+    bool has_residual) {
   check_cl(dy, "dy"); check_cl(x, "x");
   int64_t C = x.size(1);
   int64_t M = x.numel() / C;
   auto s = cur_stream();
   int dt = dt_code(x);
+  // relu mask source: with a residual, y != bn(x) so the mask must come
+  // from y; without one, recompute (w*xhat+b > 0) and skip reading y.
+  const int relu_mode = relu ? (has_residual ? 1 : 2) : 0;
   auto sums = at::empty({2 * C}, x.options().dtype(at::kFloat));
   int chunks = ddl_bn_stats_chunks(M, C);
   auto partials = at::empty({(int64_t)chunks * 2 * C}, x.options().dtype(at::kFloat));
-  ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu,
-                   mean.data_ptr<float>(), rstd.data_ptr<float>(), M, C, dt,
+  ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu_mode,
+                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                   weight.data_ptr(), bias.data_ptr(), M, C, dt,
                    sums.data_ptr<float>(), partials.data_ptr<float>(), chunks, s);
   auto dx = at::empty_like(x);
   at::Tensor dres;
@@ -282,8 +287,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
   }
   ddl_bn_bwd_apply(dy.data_ptr(), x.data_ptr(), y.data_ptr(), dx.data_ptr(),
                    dres_p, mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                   weight.data_ptr(), sums.data_ptr<float>(), M, C, relu, dt,
-                   s);
+                   weight.data_ptr(), bias.data_ptr(),
+                   sums.data_ptr<float>(), M, C, relu_mode, dt, s);
   auto dw = at::empty_like(weight);
   auto db = at::empty_like(weight);
   ddl_bn_grads_from_sums(sums.data_ptr<float>(), dw.data_ptr(), db.data_ptr(),

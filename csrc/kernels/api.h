@@ -55,17 +55,20 @@ void ddl_bn_infer(const void* x, void* y, const void* residual,
                   const float* running_mean, const float* running_var,
                   float eps, const void* weight, const void* bias, int64_t M,
                   int C, int relu, int dtype, hipStream_t s);
-// backward stats: sums[0..C-1] = sum(dy_m * xhat), sums[C..] = sum(dy_m)
-// where dy_m = dy * (y>0) if relu else dy.  Pre-zeroed.
-void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
-                      const float* mean, const float* rstd, int64_t M, int C,
+// backward stats: sums[0..C-1] = sum(dy_m * xhat), sums[C..] = sum(dy_m).
+// relu_mode: 0 none; 1 mask from y (residual); 2 mask recomputed as
+// (w*xhat+b > 0) — skips the y read (one activation pass saved).
+void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y,
+                      int relu_mode, const float* mean, const float* rstd,
+                      const void* weight, const void* bias, int64_t M, int C,
                       int dtype, float* sums, float* partials, int max_chunks,
                       hipStream_t s);
 // dx (+ optional dres copy of masked dy), dweight/dbias cast from sums.
 void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y,
                       void* dx, void* dres, const float* mean,
-                      const float* rstd, const void* weight, const float* sums,
-                      int64_t M, int C, int relu, int dtype, hipStream_t s);
+                      const float* rstd, const void* weight, const void* bias,
+                      const float* sums, int64_t M, int C, int relu_mode,
+                      int dtype, hipStream_t s);
 void ddl_bn_grads_from_sums(const float* sums, void* dweight, void* dbias,
                             int C, int dtype, hipStream_t s);
 

@@ -195,12 +195,17 @@ __global__ void bn_infer_k(const T* __restrict__ x, T* __restrict__ y,
 }
 
 // ---- backward stats: sums[c] = sum(dy_m * xhat), sums[C+c] = sum(dy_m) ----
-template <typename T, bool RELU>
+// RELU modes: 0 = no relu; 1 = mask from y (residual case — y != bn(x));
+// 2 = recompute mask as (w*xhat + b > 0), skipping the y read entirely
+// (one full activation pass saved; exact for relu-without-residual).
+template <typename T, int RELU>
 __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
                                const T* __restrict__ x,
                                const T* __restrict__ y,
                                const float* __restrict__ mean,
-                               const float* __restrict__ rstd, int64_t M,
+                               const float* __restrict__ rstd,
+                               const T* __restrict__ weight,
+                               const T* __restrict__ bias, int64_t M,
                                int C, float* __restrict__ sums) {
   __shared__ float red[2][ROWS][CG * 8];
   const int vg = threadIdx.x % CG;
@@ -208,17 +213,24 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
   const int c0 = blockIdx.x * CG * 8 + vg * 8;
   if (c0 >= C) return;
   float s0[8] = {0}, s1[8] = {0};
+  Vec8<T> w8, b8;
+  if (RELU == 2) {
+    w8 = load8(weight + c0);
+    b8 = load8(bias + c0);
+  }
   for (int64_t m = (int64_t)blockIdx.y * ROWS + trow; m < M;
        m += (int64_t)gridDim.y * ROWS) {
     Vec8<T> vdy = load8(dy + m * C + c0);
     Vec8<T> vx = load8(x + m * C + c0);
     Vec8<T> vy;
-    if (RELU) vy = load8(y + m * C + c0);
+    if (RELU == 1) vy = load8(y + m * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g = to_f32(vdy.v[j]);
-      if (RELU) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
       float xh = (to_f32(vx.v[j]) - mean[c0 + j]) * rstd[c0 + j];
+      if (RELU == 1) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
+      if (RELU == 2)
+        g = (xh * to_f32(w8.v[j]) + to_f32(b8.v[j])) > 0.f ? g : 0.f;
       s0[j] += g * xh;
       s1[j] += g;
     }
@@ -251,7 +263,8 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
 }
 
 // ---- backward apply: dx = w*rstd*(dy_m - s1/M - xhat*s0/M); dres = dy_m ---
-template <typename T, bool RELU, bool RES>
+// RELU: 0 none / 1 mask from y (residual) / 2 recomputed (w*xhat+b > 0).
+template <typename T, int RELU, bool RES>
 __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
                                const T* __restrict__ x,
                                const T* __restrict__ y, T* __restrict__ dx,
@@ -259,6 +272,7 @@ __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
                                const float* __restrict__ mean,
                                const float* __restrict__ rstd,
                                const T* __restrict__ weight,
+                               const T* __restrict__ bias,
                                const float* __restrict__ sums, int64_t M,
                                int C) {
   const int ncg = C / 8;
@@ -271,15 +285,19 @@ __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
     Vec8<T> vdy = load8(dy + m * C + c0);
     Vec8<T> vx = load8(x + m * C + c0);
     Vec8<T> w8 = load8(weight + c0);
+    Vec8<T> b8;
+    if (RELU == 2) b8 = load8(bias + c0);
     Vec8<T> vy, out, rout;
-    if (RELU) vy = load8(y + m * C + c0);
+    if (RELU == 1) vy = load8(y + m * C + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
       float g = to_f32(vdy.v[j]);
-      if (RELU) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
-      if (RES) rout.v[j] = from_f32<T>(g);
       float xh = (to_f32(vx.v[j]) - mean[c]) * rstd[c];
+      if (RELU == 1) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
+      if (RELU == 2)
+        g = (xh * to_f32(w8.v[j]) + to_f32(b8.v[j])) > 0.f ? g : 0.f;
+      if (RES) rout.v[j] = from_f32<T>(g);
       float t = g - sums[C + c] * invM - xh * sums[c] * invM;
       out.v[j] = from_f32<T>(to_f32(w8.v[j]) * rstd[c] * t);
     }
@@ -372,17 +390,20 @@ void ddl_bn_infer(const void* x, void* y, const void* residual,
 #undef RUN
 }
 
-void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
-                      const float* mean, const float* rstd, int64_t M, int C,
+void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y,
+                      int relu_mode, const float* mean, const float* rstd,
+                      const void* weight, const void* bias, int64_t M, int C,
                       int dtype, float* sums, float* partials, int max_chunks,
                       hipStream_t s) {
   dim3 g = stats_grid(M, C);
   if ((int)g.y > max_chunks) g.y = max_chunks;
 #define RUN(T, RELU)                                                           \
   bn_bwd_stats_k<T, RELU><<<g, 256, 0, s>>>(                                   \
-      (const T*)dy, (const T*)x, (const T*)y, mean, rstd, M, C, partials)
+      (const T*)dy, (const T*)x, (const T*)y, mean, rstd, (const T*)weight,    \
+      (const T*)bias, M, C, partials)
 #define SEL(T)                                                                 \
-  if (relu) RUN(T, true); else RUN(T, false)
+  if (relu_mode == 2) RUN(T, 2); else if (relu_mode == 1) RUN(T, 1);           \
+  else RUN(T, 0)
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN
@@ -391,18 +412,21 @@ void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y, int relu,
 
 void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y, void* dx,
                       void* dres, const float* mean, const float* rstd,
-                      const void* weight, const float* sums, int64_t M, int C,
-                      int relu, int dtype, hipStream_t s) {
+                      const void* weight, const void* bias, const float* sums,
+                      int64_t M, int C, int relu_mode, int dtype,
+                      hipStream_t s) {
   int g = apply_grid(M, C);
 #define RUN(T, RELU, RES)                                                      \
   bn_bwd_apply_k<T, RELU, RES><<<g, 256, 0, s>>>(                              \
       (const T*)dy, (const T*)x, (const T*)y, (T*)dx, (T*)dres, mean, rstd,    \
-      (const T*)weight, sums, M, C)
+      (const T*)weight, (const T*)bias, sums, M, C)
 #define SEL(T)                                                                 \
-  if (relu && dres) RUN(T, true, true);                                        \
-  else if (relu) RUN(T, true, false);                                          \
-  else if (dres) RUN(T, false, true);                                          \
-  else RUN(T, false, false)
+  if (relu_mode == 1 && dres) RUN(T, 1, true);                                 \
+  else if (relu_mode == 2 && dres) RUN(T, 2, true);                            \
+  else if (relu_mode == 2) RUN(T, 2, false);                                   \
+  else if (relu_mode == 1) RUN(T, 1, false);                                   \
+  else if (dres) RUN(T, 0, true);                                              \
+  else RUN(T, 0, false)
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN

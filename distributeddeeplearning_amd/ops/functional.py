@@ -124,7 +124,7 @@ class _BnAddReluFn(torch.autograd.Function):
             float(momentum), float(eps), relu,
             residual if residual is not None else None,
         )
-        ctx.save_for_backward(x, weight, mean, rstd, y)
+        ctx.save_for_backward(x, weight, bias, mean, rstd, y)
         ctx.relu = relu
         ctx.has_residual = residual is not None
         return y
@@ -132,10 +132,10 @@ class _BnAddReluFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = load_kernels_ext()
-        x, weight, mean, rstd, y = ctx.saved_tensors
+        x, weight, bias, mean, rstd, y = ctx.saved_tensors
         dy = to_channels_last(dy)
         dx, dw, db, dres = ext.bn_relu_bwd(
-            dy, x, weight, mean, rstd, y, ctx.relu, ctx.has_residual)
+            dy, x, weight, bias, mean, rstd, y, ctx.relu, ctx.has_residual)
         return (dx, dw, db, None, None, None, None, None,
                 dres if ctx.has_residual else None)
 

@@ -145,7 +145,8 @@ def test_bn_relu_residual_fwd_bwd():
                                     0.1, 1e-5, True, resg)
     assert relerr(y, ref) < 0.02
     dyg = cl(dy.to(DEV).bfloat16())
-    dx, dw, db, dres = ext.bn_relu_bwd(dyg, xg, wgt_g, mean, rstd, y, True, True)
+    dx, dw, db, dres = ext.bn_relu_bwd(dyg, xg, wgt_g, bg, mean, rstd, y,
+                                       True, True)
     assert relerr(dres, res.grad) < 0.02
     assert relerr(dw, wgt.grad) < 0.02
     assert relerr(db, b.grad) < 0.02
@@ -275,3 +276,26 @@ def test_dropout():
     assert abs(kept - 0.7) < 0.02
     nz = y[y.float() > 0].float()
     assert torch.allclose(nz, torch.full_like(nz, 1.0 / 0.7), rtol=1e-2)
+
+
+def test_bn_relu_no_residual_bwd_recomputed_mask():
+    """bn1/bn2 style: relu without residual — backward recomputes the mask
+    from w*xhat+b (never reads y)."""
+    x = mk((4, 128, 14, 14), seed=31).float().requires_grad_(True)
+    wgt = (torch.rand(128) - 0.2).requires_grad_(True)  # mixed-sign gammas
+    b = (torch.randn(128) * 0.2).requires_grad_(True)
+    rm, rv = torch.zeros(128), torch.ones(128)
+    ref = F.relu(F.batch_norm(x, rm.clone(), rv.clone(), wgt, b, True, 0.1, 1e-5))
+    dy = mk(ref.shape, seed=32).float()
+    ref.backward(dy)
+    ext = _ext()
+    xg = cl(x.detach().to(DEV).bfloat16())
+    wgt_g = wgt.detach().bfloat16().to(DEV)
+    bg = b.detach().bfloat16().to(DEV)
+    y, mean, rstd = ext.bn_relu_fwd(xg, wgt_g, bg, rm.to(DEV), rv.to(DEV),
+                                    0.1, 1e-5, True, None)
+    dx, dw, db, _ = ext.bn_relu_bwd(cl(dy.to(DEV).bfloat16()), xg, wgt_g, bg,
+                                    mean, rstd, y, True, False)
+    assert relerr(dw, wgt.grad) < 0.03
+    assert relerr(db, b.grad) < 0.03
+    assert relerr(dx, x.grad) < 0.06
