@@ -8,6 +8,7 @@
 // stores, no atomics) combined by a deterministic reduce+cast kernel.
 #include "conv_tile.h"
 #include "api.h"
+#include <stdlib.h>
 
 namespace {
 
@@ -300,8 +301,10 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   pl.ntk = (int)cdiv(K, BM);
   pl.ntc = (int)cdiv(pl.kred_total, BN);
   pl.rs_total = flat ? 1 : R * S;
+  const char* bt = getenv("DDL_WGRAD_BLOCKS");
+  const int64_t target = bt ? atoll(bt) : 1024;
   int64_t chunks = std::max<int64_t>(
-      1, 1024 / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
+      1, target / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
   chunks = std::min<int64_t>(chunks, cdiv(M, BK));
   chunks = std::min<int64_t>(chunks, 512);  // bound slab traffic
   // cap slab memory at ~256 MB
