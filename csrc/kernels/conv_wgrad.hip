@@ -302,7 +302,7 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   pl.ntc = (int)cdiv(pl.kred_total, BN);
   pl.rs_total = flat ? 1 : R * S;
   const char* bt = getenv("DDL_WGRAD_BLOCKS");
-  const int64_t target = bt ? atoll(bt) : 1024;
+  const int64_t target = bt ? atoll(bt) : 512;  // measured best (sweep 256-4096)
   int64_t chunks = std::max<int64_t>(
       1, target / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
   chunks = std::min<int64_t>(chunks, cdiv(M, BK));
