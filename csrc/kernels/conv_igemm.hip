@@ -19,7 +19,19 @@
 
 namespace {
 
-enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2 };
+enum Mode { FWD = 0, DGRAD = 1, FWD_FLAT = 2, DGRAD_S2 = 3 };
+
+// Strided-dgrad phase parameters: output rows h = hh*sh + ph2 (w alike)
+// only receive taps r = r0 + i*sh, making every gather row dense (a
+// non-phased stride-2 dgrad wastes 3/4 of its MFMA work on masked rows —
+// profiles/conv_shape_tflops.md).
+struct DgradPhase {
+  int ph2, pw2;      // this phase's (h % sh, w % sw)
+  int r0, s0;        // first valid tap per axis
+  int Sp;            // valid taps along s
+  int Hp, Wp;        // phase subgrid dims
+  FastDiv fdWp, fdHpWp;
+};
 
 __device__ __align__(256) const unsigned char ddl_zero_page[256] = {0};
 
@@ -35,7 +47,7 @@ template <typename T, int MODE, bool DBUF>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
-    int Nc, int relu) {
+    int Nc, int relu, DgradPhase gp) {
   __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
   char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
   char* const lB[2] = {smem + TILE_BYTES,
@@ -61,7 +73,14 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   for (int i = 0; i < 4; ++i) {
     int64_t m = tile_m + trow + 32 * i;
     if (m < M) {
-      if (MODE == DGRAD) {
+      if (MODE == DGRAD_S2) {
+        // m indexes the phase subgrid; store ACTUAL (h, w)
+        const uint32_t mw = fd_div((uint32_t)m, gp.fdWp);
+        a_i2[i] = (int)((uint32_t)m - mw * gp.Wp) * d.sw + gp.pw2;
+        const uint32_t n = fd_div((uint32_t)m, gp.fdHpWp);
+        a_i1[i] = (int)(mw - n * gp.Hp) * d.sh + gp.ph2;
+        a_n[i] = (int)n;
+      } else if (MODE == DGRAD) {
         a_i2[i] = (int)(m % d.W);
         a_i1[i] = (int)((m / d.W) % d.H);
         a_n[i] = (int)(m / ((int64_t)d.W * d.H));
@@ -75,11 +94,14 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
 
-  const int Cred = (MODE == DGRAD) ? d.K : d.C;
-  const int64_t a_ld = (MODE == DGRAD) ? d.K : d.C;
-  const int b_ld = (MODE == DGRAD) ? d.K : d.C;
+  const bool isdg = (MODE == DGRAD || MODE == DGRAD_S2);
+  const int Cred = isdg ? d.K : d.C;
+  const int64_t a_ld = isdg ? d.K : d.C;
+  const int b_ld = isdg ? d.K : d.C;
   constexpr int BKT = ConvElems<T>::BKT;  // reduce elements per LDS tile
-  const int rs_total = (MODE == FWD_FLAT) ? 1 : d.R * d.S;
+  const int Rp = (MODE == DGRAD_S2) ? (d.R - gp.r0 + d.sh - 1) / d.sh : d.R;
+  const int Sp = (MODE == DGRAD_S2) ? gp.Sp : d.S;
+  const int rs_total = (MODE == FWD_FLAT) ? 1 : Rp * Sp;
   const int kred_total = (MODE == FWD_FLAT) ? d.R * d.S * d.C : Cred;
   const int nct = (kred_total + BKT - 1) / BKT;
   const int total_tiles = rs_total * nct;
@@ -90,7 +112,14 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   const T* zpad = reinterpret_cast<const T*>(ddl_zero_page);
 
   auto update_rs_ptrs = [&](int rs) {
-    const int r = rs / d.S, s = rs % d.S;
+    int r, s;
+    if (MODE == DGRAD_S2) {
+      r = gp.r0 + (rs / Sp) * d.sh;
+      s = gp.s0 + (rs % Sp) * d.sw;
+    } else {
+      r = rs / d.S;
+      s = rs % d.S;
+    }
     cur_rs = rs;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
@@ -249,7 +278,6 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   };
 
-  const int c_last = (nct - 1) * BK;
   if (MODE == FWD_FLAT) {
     load_tile_flat(0);
     int c0n = 0;
@@ -263,6 +291,8 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       mfma_tile(lA[0], lB[0]);
       __syncthreads();
     }
+  } else if (total_tiles == 0) {
+    // empty phase (e.g. stride-2 1x1 odd rows): output is all zeros
   } else if (DBUF) {
     // Counted-vmcnt pipeline (guide §5 T3/T4 minimum form): raw s_barrier
     // (no implicit vmcnt(0) drain) + `s_waitcnt vmcnt(8)` lets tile t+1's
@@ -304,6 +334,18 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   }
 
   // ---- epilogue -----------------------------------------------------------
+  // For DGRAD_S2 the kernel's m-space is the phase subgrid: map each row
+  // back to its actual (n, h, w) output offset.
+  auto out_row_elem = [&](int64_t m) -> int64_t {
+    if (MODE == DGRAD_S2) {
+      const uint32_t mw = fd_div((uint32_t)m, gp.fdWp);
+      const int w_ = (int)((uint32_t)m - mw * gp.Wp) * d.sw + gp.pw2;
+      const uint32_t n_ = fd_div((uint32_t)m, gp.fdHpWp);
+      const int h_ = (int)(mw - n_ * gp.Hp) * d.sh + gp.ph2;
+      return (((int64_t)n_ * d.H + h_) * d.W + w_) * (int64_t)Nc;
+    }
+    return m * Nc;
+  };
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
   if constexpr (sizeof(T) == 4) {
@@ -312,8 +354,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     for (int nr = 0; nr < 4; ++nr) {
       const int n = tile_n + wc + nr * 16 + en;
       if (n >= Nc) continue;
-      const float bv = (MODE != DGRAD && bias != nullptr)
-                           ? to_f32(bias[n]) : 0.f;
+      const float bv = (!isdg && bias != nullptr) ? to_f32(bias[n]) : 0.f;
 #pragma unroll
       for (int mr = 0; mr < 4; ++mr) {
 #pragma unroll
@@ -322,7 +363,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
           if (m >= M) continue;
           float f = acc[mr][nr][reg] + bv;
           if (relu) f = fmaxf(f, 0.f);
-          out[m * Nc + n] = f;
+          out[out_row_elem(m) + n] = f;
         }
       }
     }
@@ -334,7 +375,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
 #pragma unroll
   for (int nr = 0; nr < 4; ++nr) {
     const int n_loc = wc + nr * 16 + en;
-    const float bv = (MODE != DGRAD && bias != nullptr && tile_n + n_loc < Nc)
+    const float bv = (!isdg && bias != nullptr && tile_n + n_loc < Nc)
                          ? to_f32(bias[tile_n + n_loc]) : 0.f;
 #pragma unroll
     for (int mr = 0; mr < 4; ++mr) {
@@ -355,7 +396,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     const int ub = (unit & 15) * 8;
     const int64_t m = tile_m + row;
     if (m < M && ub < nvalid) {
-      *reinterpret_cast<uint4*>(out + m * Nc + tile_n + ub) =
+      *reinterpret_cast<uint4*>(out + out_row_elem(m) + tile_n + ub) =
           *reinterpret_cast<const uint4*>(etile + row * BN + ub);
     }
   }
@@ -382,9 +423,10 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
   const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
   const bool dbuf = conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
+  DgradPhase gp{};  // unused by FWD modes
 #define RUN(T, MODE, DB)                                                       \
   conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
-      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu)
+      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp)
 #define SEL(T, MODE)                                                           \
   if (dbuf) RUN(T, MODE, true); else RUN(T, MODE, false)
   if (dtype == 0) { if (flat) { SEL(float, FWD_FLAT); } else { SEL(float, FWD); } }
@@ -398,17 +440,45 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
                     int W, int C, int K, int R, int S, int P, int Q, int sh,
                     int sw, int ph, int pw, int dtype, hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
-  const int64_t M = (int64_t)N * H * W;
   const int Nc = C;
-  const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
   const bool dbuf = conv_use_dbuf(K, R * S);
+  if (sh == 1 && sw == 1) {
+    const int64_t M = (int64_t)N * H * W;
+    const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+    DgradPhase gp{};
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0)
-  if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
-  else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
-  else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp)
+    if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
+    else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
+    else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
 #undef RUN
+    return;
+  }
+  // strided dgrad: one dense launch per (h%sh, w%sw) phase
+  for (int ph2 = 0; ph2 < sh; ++ph2) {
+    for (int pw2 = 0; pw2 < sw; ++pw2) {
+      DgradPhase gp;
+      gp.ph2 = ph2; gp.pw2 = pw2;
+      gp.r0 = ((ph2 + ph) % sh + sh) % sh;
+      gp.s0 = ((pw2 + pw) % sw + sw) % sw;
+      gp.Sp = gp.s0 < S ? (S - gp.s0 + sw - 1) / sw : 0;
+      gp.Hp = ph2 < H ? (H - ph2 + sh - 1) / sh : 0;
+      gp.Wp = pw2 < W ? (W - pw2 + sw - 1) / sw : 0;
+      if (gp.Hp == 0 || gp.Wp == 0) continue;
+      gp.fdWp = make_fastdiv((uint32_t)gp.Wp);
+      gp.fdHpWp = make_fastdiv((uint32_t)gp.Hp * gp.Wp);
+      const int64_t M = (int64_t)N * gp.Hp * gp.Wp;
+      const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+#define RUN(T, DB)                                                             \
+  conv_igemm_k<T, DGRAD_S2, DB><<<(unsigned)grid, 256, 0, s>>>(                \
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp)
+      if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
+      else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
+      else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
+#undef RUN
+    }
+  }
 }
 
 }  // extern "C"
