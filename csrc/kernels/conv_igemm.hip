@@ -130,6 +130,15 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
           const int wi = a_i2[i] * d.sw + s - d.pw;
           if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
             a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.H + hi) * d.W + wi) * a_ld;
+        } else if (MODE == DGRAD_S2) {
+          // phase construction guarantees hn % sh == 0 and wn % sw == 0
+          const int hn = a_i1[i] + d.ph - r;
+          const int wn = a_i2[i] + d.pw - s;
+          if (hn >= 0 && wn >= 0) {
+            const int ho = hn / d.sh, wo = wn / d.sw;
+            if (ho < d.P && wo < d.Q)
+              a_ptr[i] = Asrc + (((int64_t)a_n[i] * d.P + ho) * d.Q + wo) * a_ld;
+          }
         } else if (MODE == DGRAD) {
           const int hn = a_i1[i] + d.ph - r;
           const int wn = a_i2[i] + d.pw - s;
