@@ -30,9 +30,12 @@ at::Tensor empty_nhwc(int64_t n, int64_t c, int64_t h, int64_t w,
 }
 
 void check_cl(const at::Tensor& t, const char* name) {
+  // Strict: a plain NCHW-contiguous activation would silently read as
+  // garbage NHWC.  The ambiguous cases (C==1 or 1x1 spatial) are the only
+  // ones where NCHW- and NHWC-contiguity coincide.
   TORCH_CHECK(t.is_contiguous(at::MemoryFormat::ChannelsLast) ||
-                  (t.dim() == 4 && (t.size(1) == 1 || (t.size(2) == 1 && t.size(3) == 1))) ||
-                  t.is_contiguous(),
+                  (t.dim() == 4 &&
+                   (t.size(1) == 1 || (t.size(2) == 1 && t.size(3) == 1))),
               name, " must be channels_last contiguous");
 }
 

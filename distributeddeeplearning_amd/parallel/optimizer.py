@@ -175,6 +175,10 @@ class DistributedOptimizer:
 
     def zero_grad(self, set_to_none: bool = False):
         # Grads are views into flat buckets; zero the buckets and keep views.
+        # Mid-accumulation (backward_passes_per_step > 1, before the real
+        # step) the call is a no-op so accumulated gradients survive.
+        if self._passes != 0:
+            return
         self.buckets.zero_grads()
         self.buckets.reset()
         if not self.buckets.grad_views_valid():

@@ -88,3 +88,35 @@ def test_set_to_none_guard():
     opt.zero_grad()
     m(torch.randn(2, 64)).sum().backward()
     opt.step()
+
+
+def test_grad_accumulation_sums_both_passes():
+    """2-pass accumulation must apply g1+g2, and zero_grad between passes
+    must not wipe the accumulated gradients."""
+    import copy
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("single")
+    torch.manual_seed(11)
+    m1 = torch.nn.Linear(8, 8)
+    m2 = copy.deepcopy(m1)
+
+    o1 = DistributedOptimizer(FusedSGD(m1.parameters(), lr=0.1),
+                              named_parameters=m1.named_parameters(),
+                              backward_passes_per_step=2)
+    x1, x2 = torch.randn(4, 8), torch.randn(4, 8)
+    m1(x1).sum().backward()
+    o1.step()        # accumulation pass: no update
+    o1.zero_grad()   # must be a no-op mid-accumulation
+    m1(x2).sum().backward()
+    o1.step()        # real step on g(x1)+g(x2)
+    o1.zero_grad()
+
+    o2 = DistributedOptimizer(FusedSGD(m2.parameters(), lr=0.1),
+                              named_parameters=m2.named_parameters())
+    m2(x1).sum().backward()
+    m2(x2).sum().backward()  # same accumulated grads in one pass
+    o2.step()
+
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-7)

@@ -189,3 +189,33 @@ def _w_compression(rank, world):
 def test_fp16_compression_keeps_ranks_synced():
     res = run_distributed(_w_compression, world=2)
     assert torch.equal(res[0]["params"], res[1]["params"])
+
+
+def _w_world4(rank, world):
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("torchdist")
+    model = _make_mlp(seed=9)
+    api.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = DistributedOptimizer(
+        FusedSGD(model.parameters(), lr=0.05, momentum=0.9),
+        named_parameters=model.named_parameters(), bucket_bytes=2048,
+        backward_passes_per_step=2)
+    for step in range(4):  # 2 optimizer steps with 2-pass accumulation
+        g = torch.Generator().manual_seed(step)
+        x = torch.randn(8, 16, generator=g)
+        y = torch.randn(8, 4, generator=g)
+        xr = x[rank * 2:(rank + 1) * 2]
+        yr = y[rank * 2:(rank + 1) * 2]
+        torch.nn.functional.mse_loss(model(xr), yr).backward()
+        opt.step()
+        opt.zero_grad()
+    params = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    return {"params": params}
+
+
+def test_world4_with_grad_accumulation():
+    """4 ranks x 2-pass accumulation over gloo: all ranks stay identical."""
+    res = run_distributed(_w_world4, world=4, timeout=300)
+    for r in res[1:]:
+        assert torch.equal(res[0]["params"], r["params"])
