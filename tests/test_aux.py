@@ -133,3 +133,16 @@ def test_tune_buckets_cpu_smoke(tmp_path):
         env={**os.environ, "DDL_COMM": "single"})
     assert r.returncode == 0, r.stderr[-800:]
     assert "best_bucket_mb" in r.stdout
+
+
+def test_example_script_cpu_smoke(tmp_path):
+    """examples/imagenet_resnet50.py (the reference-script port) runs
+    end-to-end on CPU with synthetic data and writes a checkpoint."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "imagenet_resnet50.py"),
+         "--epochs", "1", "--steps-per-epoch", "2", "--batch-size", "4",
+         "--checkpoint-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=600,
+        env={**os.environ, "DDL_COMM": "single"})
+    assert r.returncode == 0, r.stderr[-800:]
+    assert (tmp_path / "resnet50_epoch1.pt").exists()
