@@ -16,101 +16,119 @@ enum WMode { TILED = 0, FLAT = 1 };
 
 // SIMPLE = 1x1/stride-1/no-pad conv: the x operand is just row m of the
 // input (no im2col decode; the per-tile q/p/n divides were a hot cost).
-template <typename T, int MODE, bool SIMPLE, bool DBUF>
-__global__ __launch_bounds__(256) void conv_wgrad_k(
+// TBM/TBN/NTHR: tile geometry.  (128,128,256) is the full 4-wave 2x2 tile;
+// (64,64,64) is a single-wave tile for small K / small C*R*S convs (layer1
+// of ResNet: a 128^2 tile on a 64^2 dw plane wastes 75% of the MFMA work —
+// the small tile instead fills the chip with extra split-K chunks).
+// The 64^2 TILED/SIMPLE variants sit at ~200 VGPR, just past the 192
+// 2-waves/SIMD boundary; minWavesPerEU=2 squeezes them back under it
+// (FLAT's im2col decode needs the registers more than the occupancy).
+template <typename T, int MODE, bool SIMPLE, bool DBUF, int TBM = 128,
+          int TBN = 128, int NTHR = 256>
+__global__ __launch_bounds__(NTHR, (NTHR == 64) ? 2 : 1)
+void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
     float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows,
     int64_t slab_len, FastDiv fdq, FastDiv fdpq) {
+  static_assert(TBM == TBN, "square tiles only");
+  constexpr int ATILE = TBM * 128;  // bytes per operand tile (128 B rows)
   // each m-chunk (blockIdx.z) owns a private fp32 slab: plain dense stores,
   // no atomics (a chunks-deep atomicAdd chain per dw element serializes at
   // L2 like the BN-stats case), and the final sum is deterministic.
   dw_ws += (int64_t)blockIdx.z * slab_len;
-  __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
-  char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
-  char* const lB[2] = {smem + TILE_BYTES,
-                       DBUF ? smem + 3 * TILE_BYTES : smem + TILE_BYTES};
+  __shared__ __align__(16) char smem[DBUF ? 4 * ATILE : 2 * ATILE];
+  char* const lA[2] = {smem, DBUF ? smem + 2 * ATILE : smem};
+  char* const lB[2] = {smem + ATILE,
+                       DBUF ? smem + 3 * ATILE : smem + ATILE};
 
   const int kred_total = (MODE == FLAT) ? d.R * d.S * d.C : d.C;
-  const int ntc = (kred_total + BN - 1) / BN;
-  const int tile_k = (int)(blockIdx.x / ntc) * BM;
-  const int tile_c = (int)(blockIdx.x % ntc) * BN;
+  const int ntc = (kred_total + TBN - 1) / TBN;
+  const int tile_k = (int)(blockIdx.x / ntc) * TBM;
+  const int tile_c = (int)(blockIdx.x % ntc) * TBN;
   const int rs = (MODE == FLAT) ? 0 : blockIdx.y;
   const int r = rs / d.S, s = rs % d.S;
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wr = (wid >> 1) * 64;
-  const int wc = (wid & 1) * 64;
+  const int wr = (NTHR == 64) ? 0 : (wid >> 1) * 64;
+  const int wc = (NTHR == 64) ? 0 : (wid & 1) * 64;
   // staging assignment (in-register transpose): each thread owns a
-  // 16B-wide k/c group x 4 consecutive m.  2-byte dtypes: 16 groups of 8
-  // rows x 16 m-quads -> 8 b64 transposed writes; fp32: 32 groups of 4
-  // rows x 8 m-quads -> 4 b128 transposed writes.
+  // 16B-wide k/c group x 4 consecutive m, in PASSES m-column passes.
+  // (128,256) 2-byte dtypes: 16 groups of 8 rows x 16 m-quads, 1 pass;
+  // (64,64) 2-byte: 8 groups x 8 m-quads x 2 passes; fp32 analogous.
   constexpr int EPU = ConvElems<T>::PER_UNIT;   // k rows per 16 B load
-  constexpr int NSU = 256 / EPU / 2;            // staging groups (16 / 32)
+  constexpr int NSU = TBM / EPU;                // staging k/c groups
+  constexpr int BKT = ConvElems<T>::BKT;        // m columns per LDS tile
+  constexpr int MPP = (NTHR / NSU) * 4;         // m columns per pass
+  constexpr int PASSES = BKT / MPP;
+  static_assert(PASSES >= 1 && BKT % MPP == 0, "staging geometry");
   const int su = threadIdx.x & (NSU - 1);       // k/c group
   const int sm4 = (threadIdx.x / NSU) * 4;      // first of 4 owned m columns
 
   f32x4_t acc[4][4] = {};
 
-  constexpr int BKT = ConvElems<T>::BKT;  // m columns per LDS tile (64 / 32)
   const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
   const int64_t m_end = min(M, m_begin + chunk_rows);
   const int64_t ntiles =
       (m_end > m_begin) ? (m_end - m_begin + BKT - 1) / BKT : 0;
 
-  uint4 va[4], vb[4];
+  uint4 va[PASSES][4], vb[PASSES][4];
   auto load_tile = [&](int64_t ti) {
     const int64_t m0 = m_begin + ti * BKT;
     const int k0g = tile_k + su * EPU;
 #pragma unroll
-    for (int mm = 0; mm < 4; ++mm) {
-      const int64_t m = m0 + sm4 + mm;
-      va[mm] = zero16();
-      if (m < m_end && k0g + EPU <= d.K)
-        va[mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
-    }
+    for (int ps = 0; ps < PASSES; ++ps)
 #pragma unroll
-    for (int mm = 0; mm < 4; ++mm) {
-      const int64_t m = m0 + sm4 + mm;
-      vb[mm] = zero16();
-      T* e = reinterpret_cast<T*>(&vb[mm]);
-      if (SIMPLE) {
-        const int c0 = tile_c + su * EPU;
-        if (m < m_end && c0 + EPU <= d.C)
-          vb[mm] = *reinterpret_cast<const uint4*>(x + m * d.C + c0);
-      } else if (m < m_end) {
-        const uint32_t mq = fd_div((uint32_t)m, fdq);
-        const int q = (int)((uint32_t)m - mq * d.Q);
-        const uint32_t n = fd_div((uint32_t)m, fdpq);
-        const int p = (int)(mq - n * d.P);
-        if (MODE == TILED) {
-          const int hi = p * d.sh + r - d.ph;
-          const int wi = q * d.sw + s - d.pw;
+      for (int mm = 0; mm < 4; ++mm) {
+        const int64_t m = m0 + ps * MPP + sm4 + mm;
+        va[ps][mm] = zero16();
+        if (m < m_end && k0g + EPU <= d.K)
+          va[ps][mm] = *reinterpret_cast<const uint4*>(dy + m * d.K + k0g);
+      }
+#pragma unroll
+    for (int ps = 0; ps < PASSES; ++ps)
+#pragma unroll
+      for (int mm = 0; mm < 4; ++mm) {
+        const int64_t m = m0 + ps * MPP + sm4 + mm;
+        vb[ps][mm] = zero16();
+        T* e = reinterpret_cast<T*>(&vb[ps][mm]);
+        if (SIMPLE) {
           const int c0 = tile_c + su * EPU;
-          if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + EPU <= d.C)
-            vb[mm] = *reinterpret_cast<const uint4*>(
-                x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
-        } else {
+          if (m < m_end && c0 + EPU <= d.C)
+            vb[ps][mm] = *reinterpret_cast<const uint4*>(x + m * d.C + c0);
+        } else if (m < m_end) {
+          const uint32_t mq = fd_div((uint32_t)m, fdq);
+          const int q = (int)((uint32_t)m - mq * d.Q);
+          const uint32_t n = fd_div((uint32_t)m, fdpq);
+          const int p = (int)(mq - n * d.P);
+          if (MODE == TILED) {
+            const int hi = p * d.sh + r - d.ph;
+            const int wi = q * d.sw + s - d.pw;
+            const int c0 = tile_c + su * EPU;
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W && c0 + EPU <= d.C)
+              vb[ps][mm] = *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)n * d.H + hi) * d.W + wi) * d.C + c0);
+          } else {
 #pragma unroll
-          for (int j = 0; j < EPU; ++j) {
-            const int kred = tile_c + su * EPU + j;
-            if (kred < kred_total) {
-              const int rr = kred / (d.S * d.C);
-              const int rem = kred % (d.S * d.C);
-              const int ss = rem / d.C;
-              const int cc = rem % d.C;
-              const int hi = p * d.sh + rr - d.ph;
-              const int wi = q * d.sw + ss - d.pw;
-              if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                e[j] = x[(((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc];
+            for (int j = 0; j < EPU; ++j) {
+              const int kred = tile_c + su * EPU + j;
+              if (kred < kred_total) {
+                const int rr = kred / (d.S * d.C);
+                const int rem = kred % (d.S * d.C);
+                const int ss = rem / d.C;
+                const int cc = rem % d.C;
+                const int hi = p * d.sh + rr - d.ph;
+                const int wi = q * d.sw + ss - d.pw;
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                  e[j] = x[(((int64_t)n * d.H + hi) * d.W + wi) * d.C + cc];
+              }
             }
           }
         }
       }
-    }
   };
 
-  auto transpose_write = [&](char* lds, const uint4 (&v)[4]) {
+  auto transpose_write = [&](char* lds, const uint4 (&v)[4], int sm4p) {
     if constexpr (sizeof(T) == 2) {
       const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
       const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
@@ -120,7 +138,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
       for (int j = 0; j < 8; ++j) {
         uint64_t pk = (uint64_t)e0[j] | ((uint64_t)e1[j] << 16) |
                       ((uint64_t)e2[j] << 32) | ((uint64_t)e3[j] << 48);
-        *reinterpret_cast<uint64_t*>(lds + swz_off(su * 8 + j, sm4 * 2)) = pk;
+        *reinterpret_cast<uint64_t*>(lds + swz_off(su * 8 + j, sm4p * 2)) = pk;
       }
     } else {
       const float* e0 = reinterpret_cast<const float*>(&v[0]);
@@ -130,13 +148,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         float4 pk{e0[j], e1[j], e2[j], e3[j]};
-        *reinterpret_cast<float4*>(lds + swz_off(su * 4 + j, sm4 * 4)) = pk;
+        *reinterpret_cast<float4*>(lds + swz_off(su * 4 + j, sm4p * 4)) = pk;
       }
     }
   };
   auto write_tile = [&](char* la, char* lb) {
-    transpose_write(la, va);
-    transpose_write(lb, vb);
+#pragma unroll
+    for (int ps = 0; ps < PASSES; ++ps) {
+      transpose_write(la, va[ps], sm4 + ps * MPP);
+      transpose_write(lb, vb[ps], sm4 + ps * MPP);
+    }
   };
 
   const int frow = lane & 15;
@@ -214,10 +235,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
   float* etile = reinterpret_cast<float*>(smem);
   const int en = lane & 15;
   const int em = (lane >> 4) * 4;
-  const int halves = DBUF ? 1 : 2;   // SBUF: two passes of 64 k-rows
+  // SBUF 128^2: two passes of 64 k-rows (etile is half the tile); the
+  // 64^2 tile's etile fits the staging smem whole.
+  constexpr int halves = (DBUF || 2 * ATILE >= TBM * TBN * 4) ? 1 : 2;
   for (int h = 0; h < halves; ++h) {
     __syncthreads();
-    const int rows_per = BM / halves;
+    const int rows_per = TBM / halves;
 #pragma unroll
     for (int mr = 0; mr < 4; ++mr) {
 #pragma unroll
@@ -226,27 +249,27 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
         if (krow < h * rows_per || krow >= (h + 1) * rows_per) continue;
 #pragma unroll
         for (int nr = 0; nr < 4; ++nr) {
-          etile[(krow - h * rows_per) * BN + wc + nr * 16 + en] =
+          etile[(krow - h * rows_per) * TBN + wc + nr * 16 + en] =
               acc[mr][nr][reg];
         }
       }
     }
     __syncthreads();
-    const int nunits = rows_per * BN / 4 / 256;  // 16B units per thread
+    const int nunits = rows_per * TBN / 4 / NTHR;  // 16B units per thread
     for (int it = 0; it < nunits; ++it) {
-      const int unit = threadIdx.x + it * 256;
-      const int row = unit >> 5;
-      const int ub = (unit & 31) * 4;
+      const int unit = (int)threadIdx.x + it * NTHR;
+      const int row = unit / (TBN / 4);
+      const int ub = (unit % (TBN / 4)) * 4;
       const int k = tile_k + h * rows_per + row;
       if (k < d.K && ub < nvalid) {
         const int cidx = tile_c + ub;
         if (MODE == FLAT) {
           for (int j = 0; j < 4 && cidx + j < kred_total; ++j)
-            dw_ws[(int64_t)k * rsc + cidx + j] = etile[row * BN + ub + j];
+            dw_ws[(int64_t)k * rsc + cidx + j] = etile[row * TBN + ub + j];
         } else {
           const int64_t o = (int64_t)k * rsc + (int64_t)rs * d.C + cidx;
           *reinterpret_cast<float4*>(dw_ws + o) =
-              *reinterpret_cast<const float4*>(etile + row * BN + ub);
+              *reinterpret_cast<const float4*>(etile + row * TBN + ub);
         }
       }
     }
@@ -254,19 +277,20 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(
 }
 
 // dw[i] = cast(sum_ch ws[ch][i]) — the deterministic chunk combine.
-// Block = 64 float4-element-groups x 4 chunk-quarters, so the chunk loop is
-// 4-way parallel and the grid covers the dw plane densely even when it is
-// small (a serial per-thread chunk loop left 95% of the chip idle).
+// Block = 16 float4-element-groups x 16 chunk-slices, so the chunk loop is
+// 16-way parallel: the 64^2-tile wgrad path runs ~2k chunks over a tiny dw
+// plane (4k elements for a 64->64 1x1), where a 4-way split left the chip
+// nearly idle.  The split order is fixed, so the sum stays deterministic.
 template <typename T>
 __global__ void wgrad_reduce_cast_k(const float* __restrict__ ws, T* dw,
                                     int64_t len, int chunks) {
-  __shared__ float red[4][64][4];
-  const int eg = threadIdx.x & 63;   // element group (4 floats)
-  const int cq = threadIdx.x >> 6;   // chunk quarter
-  const int64_t e0 = ((int64_t)blockIdx.x * 64 + eg) * 4;
+  __shared__ float red[16][16][4];
+  const int eg = threadIdx.x & 15;   // element group (4 floats)
+  const int cq = threadIdx.x >> 4;   // chunk slice
+  const int64_t e0 = ((int64_t)blockIdx.x * 16 + eg) * 4;
   float acc[4] = {0, 0, 0, 0};
   if (e0 < len) {
-    for (int ch = cq; ch < chunks; ch += 4) {
+    for (int ch = cq; ch < chunks; ch += 16) {
       const float* p = ws + (int64_t)ch * len + e0;
       if (e0 + 4 <= len) {
         float4 v = *reinterpret_cast<const float4*>(p);
@@ -281,8 +305,9 @@ __global__ void wgrad_reduce_cast_k(const float* __restrict__ ws, T* dw,
   __syncthreads();
   if (cq == 0 && e0 < len) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
-      acc[j] = red[0][eg][j] + red[1][eg][j] + red[2][eg][j] + red[3][eg][j];
+    for (int sl = 1; sl < 16; ++sl)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[j] += red[sl][eg][j];
     const int nt = (int)min((int64_t)4, len - e0);
     for (int j = 0; j < nt; ++j) dw[e0 + j] = from_f32<T>(acc[j]);
   }
@@ -290,7 +315,7 @@ __global__ void wgrad_reduce_cast_k(const float* __restrict__ ws, T* dw,
 
 struct WgradPlan {
   int64_t chunks, chunk_rows;
-  int ntk, ntc, rs_total, kred_total;
+  int ntk, ntc, rs_total, kred_total, tile;
 };
 
 WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
@@ -298,15 +323,26 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   const int64_t M = (int64_t)N * P * Q;
   const bool flat = (C % 8) != 0;
   pl.kred_total = flat ? R * S * C : C;
-  pl.ntk = (int)cdiv(K, BM);
-  pl.ntc = (int)cdiv(pl.kred_total, BN);
+  // 64^2 single-wave tile when BOTH dw-plane sides are <= 64 (ResNet
+  // layer1 64->64 convs): the 128^2 tile would run 75% dead MFMA work.
+  // One-sided-small shapes (64->256 etc.) measure FASTER on the 128^2
+  // tile: halving the tile there doubles the re-read of the streamed
+  // operand, and these kernels are traffic-bound, not MFMA-bound.  FLAT
+  // also stays on 128^2 — its im2col decode spills at 64-tile pressure.
+  pl.tile = (!flat && K <= 64 && pl.kred_total <= 64) ? 64 : 128;
+  pl.ntk = (int)cdiv(K, pl.tile);
+  pl.ntc = (int)cdiv(pl.kred_total, pl.tile);
   pl.rs_total = flat ? 1 : R * S;
-  const char* bt = getenv("DDL_WGRAD_BLOCKS");
-  const int64_t target = bt ? atoll(bt) : 512;  // measured best (sweep 256-4096)
+  // block target: the 64^2 tile runs one wave per 64-thread block, so it
+  // needs ~8x the blocks of the 4-wave 128^2 tile for the same occupancy.
+  const char* bt = getenv(pl.tile == 64 ? "DDL_WGRAD_BLOCKS64"
+                                        : "DDL_WGRAD_BLOCKS");
+  const int64_t target =
+      bt ? atoll(bt) : (pl.tile == 64 ? 2048 : 512);  // measured (swept)
   int64_t chunks = std::max<int64_t>(
       1, target / ((int64_t)pl.ntk * pl.ntc * pl.rs_total));
   chunks = std::min<int64_t>(chunks, cdiv(M, BK));
-  chunks = std::min<int64_t>(chunks, 512);  // bound slab traffic
+  chunks = std::min<int64_t>(chunks, pl.tile == 64 ? 2048 : 512);
   // cap slab memory at ~256 MB
   const int64_t slab = (int64_t)K * R * S * C * 4;
   if (slab > 0) chunks = std::min<int64_t>(chunks, std::max<int64_t>(1, (256LL << 20) / slab));
@@ -331,7 +367,8 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
                     hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int64_t M = (int64_t)N * P * Q;
-  const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
+  // must match wgrad_plan's rule (grid/mode are derived from the plan)
+  const bool flat = (C % 8) != 0;
   WgradPlan pl = wgrad_plan(N, C, K, R, S, P, Q);
   const int64_t slab_len = (int64_t)K * R * S * C;
   dim3 grid((unsigned)(pl.ntk * pl.ntc), (unsigned)pl.rs_total,
@@ -346,8 +383,14 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   conv_wgrad_k<T, MODE, SIMP, DB><<<grid, 256, 0, s>>>(                        \
       (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
       fdpq)
+#define RUN64(T, MODE, SIMP)                                                   \
+  conv_wgrad_k<T, MODE, SIMP, false, 64, 64, 64><<<grid, 64, 0, s>>>(          \
+      (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
+      fdpq)
 #define SEL(T, MODE, SIMP)                                                     \
-  if (dbuf) RUN(T, MODE, SIMP, true); else RUN(T, MODE, SIMP, false)
+  if (pl.tile == 64) RUN64(T, MODE, SIMP);                                     \
+  else if (dbuf) RUN(T, MODE, SIMP, true);                                     \
+  else RUN(T, MODE, SIMP, false)
   if (dtype == 0) {
     if (flat) { SEL(float, FLAT, false); }
     else if (simple) { SEL(float, TILED, true); }
@@ -362,8 +405,9 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
     else { SEL(f16, TILED, false); }
   }
 #undef SEL
+#undef RUN64
 #undef RUN
-  int g = (int)cdiv(slab_len, 256);
+  int g = (int)cdiv(slab_len, 64);
 #define RED(T)                                                                 \
   wgrad_reduce_cast_k<T><<<g, 256, 0, s>>>(dw_ws, (T*)dw, slab_len,            \
                                            (int)pl.chunks)
