@@ -29,13 +29,32 @@ __global__ __launch_bounds__(NTHR, (NTHR == 64) ? 2 : 1)
 void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
     float* __restrict__ dw_ws, ConvDims d, int64_t M, int64_t chunk_rows,
-    int64_t slab_len, FastDiv fdq, FastDiv fdpq) {
+    int64_t slab_len, FastDiv fdq, FastDiv fdpq, int ntiles, int chunks,
+    int xcd_map) {
   static_assert(TBM == TBN, "square tiles only");
   constexpr int ATILE = TBM * 128;  // bytes per operand tile (128 B rows)
-  // each m-chunk (blockIdx.z) owns a private fp32 slab: plain dense stores,
-  // no atomics (a chunks-deep atomicAdd chain per dw element serializes at
-  // L2 like the BN-stats case), and the final sum is deterministic.
-  dw_ws += (int64_t)blockIdx.z * slab_len;
+  // (tile, chunk) decode.  xcd_map=1: co-locate a chunk's tiles on one XCD
+  // (block b runs on XCD b%8, MI355X_MICROARCH.md §Workgroup dispatch —
+  // affinity used for speed only).  The tiles of one chunk sweep the same
+  // 64-row m-window of dy/x in near-lockstep; co-located, that window is
+  // read from HBM once per chunk instead of once per XCD that happens to
+  // host a tile (per-XCD L2s are private) — wgrad re-reads dy ntc times
+  // and x ntk times otherwise.  Used only when chunks >= 8 so all XCDs
+  // stay loaded; xcd_map=0 is the plain (x=tile, z=chunk) grid.
+  int tilei, chunk;
+  if (xcd_map) {
+    const int slot = (int)blockIdx.x >> 3;
+    chunk = ((int)blockIdx.x & 7) + 8 * (slot / ntiles);
+    tilei = slot % ntiles;
+    if (chunk >= chunks) return;
+  } else {
+    tilei = blockIdx.x;
+    chunk = blockIdx.z;
+  }
+  // each m-chunk owns a private fp32 slab: plain dense stores, no atomics
+  // (a chunks-deep atomicAdd chain per dw element serializes at L2 like
+  // the BN-stats case), and the final sum is deterministic.
+  dw_ws += (int64_t)chunk * slab_len;
   __shared__ __align__(16) char smem[DBUF ? 4 * ATILE : 2 * ATILE];
   char* const lA[2] = {smem, DBUF ? smem + 2 * ATILE : smem};
   char* const lB[2] = {smem + ATILE,
@@ -43,8 +62,8 @@ void conv_wgrad_k(
 
   const int kred_total = (MODE == FLAT) ? d.R * d.S * d.C : d.C;
   const int ntc = (kred_total + TBN - 1) / TBN;
-  const int tile_k = (int)(blockIdx.x / ntc) * TBM;
-  const int tile_c = (int)(blockIdx.x % ntc) * TBN;
+  const int tile_k = (tilei / ntc) * TBM;
+  const int tile_c = (tilei % ntc) * TBN;
   const int rs = (MODE == FLAT) ? 0 : blockIdx.y;
   const int r = rs / d.S, s = rs % d.S;
 
@@ -67,9 +86,9 @@ void conv_wgrad_k(
 
   f32x4_t acc[4][4] = {};
 
-  const int64_t m_begin = (int64_t)blockIdx.z * chunk_rows;
+  const int64_t m_begin = (int64_t)chunk * chunk_rows;
   const int64_t m_end = min(M, m_begin + chunk_rows);
-  const int64_t ntiles =
+  const int64_t mtiles =
       (m_end > m_begin) ? (m_end - m_begin + BKT - 1) / BKT : 0;
 
   uint4 va[PASSES][4], vb[PASSES][4];
@@ -203,13 +222,13 @@ void conv_wgrad_k(
     }
   };
 
-  if (ntiles > 0) {
+  if (mtiles > 0) {
     if (DBUF) {
       load_tile(0);
       write_tile(lA[0], lB[0]);
       __syncthreads();
-      for (int64_t t = 0; t < ntiles; ++t) {
-        const bool more = (t + 1) < ntiles;
+      for (int64_t t = 0; t < mtiles; ++t) {
+        const bool more = (t + 1) < mtiles;
         const int cur = (int)(t & 1);
         if (more) load_tile(t + 1);
         mfma_tile(cur ? lA[1] : lA[0], cur ? lB[1] : lB[0]);
@@ -218,10 +237,10 @@ void conv_wgrad_k(
       }
     } else {
       load_tile(0);
-      for (int64_t t = 0; t < ntiles; ++t) {
+      for (int64_t t = 0; t < mtiles; ++t) {
         write_tile(lA[0], lB[0]);
         __syncthreads();
-        if (t + 1 < ntiles) load_tile(t + 1);
+        if (t + 1 < mtiles) load_tile(t + 1);
         mfma_tile(lA[0], lB[0]);
         __syncthreads();
       }
@@ -346,8 +365,20 @@ WgradPlan wgrad_plan(int N, int C, int K, int R, int S, int P, int Q) {
   // cap slab memory at ~256 MB
   const int64_t slab = (int64_t)K * R * S * C * 4;
   if (slab > 0) chunks = std::min<int64_t>(chunks, std::max<int64_t>(1, (256LL << 20) / slab));
+  // round DOWN to a multiple of 8 so the XCD-affinity block map (chunk%8
+  // -> XCD) is load-balanced — but only when that keeps >=75% of the
+  // block target (rounding 14->8 idled half the chip on l3.c2; rounding
+  // UP past the resident-block capacity added a nearly-empty second
+  // dispatch round that doubled l1.c2's wall time).  Unrounded counts
+  // simply run the plain grid (launch checks chunks%8).
+  if (chunks >= 8) {
+    const int64_t dn = (chunks / 8) * 8;
+    if (dn * 4 >= chunks * 3) chunks = dn;
+  }
   pl.chunk_rows = cdiv(cdiv(M, chunks), BK) * BK;
-  pl.chunks = cdiv(M, pl.chunk_rows);
+  // keep the rounded count authoritative: trailing chunks may own zero
+  // rows (their blocks write zero slabs, which the reduce sums away)
+  pl.chunks = chunks;
   return pl;
 }
 
@@ -371,8 +402,21 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   const bool flat = (C % 8) != 0;
   WgradPlan pl = wgrad_plan(N, C, K, R, S, P, Q);
   const int64_t slab_len = (int64_t)K * R * S * C;
-  dim3 grid((unsigned)(pl.ntk * pl.ntc), (unsigned)pl.rs_total,
-            (unsigned)pl.chunks);
+  const int ntiles = pl.ntk * pl.ntc;
+  const char* xe = getenv("DDL_WGRAD_XCD");
+  // balanced only when chunks is a multiple of 8 (plan rounds, but the
+  // final cdiv can land one short) — otherwise use the plain grid
+  const int xcd_map =
+      (xe ? xe[0] == '1' : true) && pl.chunks >= 8 && pl.chunks % 8 == 0
+          ? 1 : 0;
+  dim3 grid;
+  if (xcd_map) {
+    grid = dim3((unsigned)(8 * cdiv(pl.chunks, 8) * ntiles),
+                (unsigned)pl.rs_total, 1u);
+  } else {
+    grid = dim3((unsigned)ntiles, (unsigned)pl.rs_total,
+                (unsigned)pl.chunks);
+  }
   const bool simple = (R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
                        pw == 0 && !flat);
   const char* env = getenv("DDL_WGRAD_DBUF");
@@ -382,11 +426,11 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
 #define RUN(T, MODE, SIMP, DB)                                                 \
   conv_wgrad_k<T, MODE, SIMP, DB><<<grid, 256, 0, s>>>(                        \
       (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
-      fdpq)
+      fdpq, ntiles, (int)pl.chunks, xcd_map)
 #define RUN64(T, MODE, SIMP)                                                   \
   conv_wgrad_k<T, MODE, SIMP, false, 64, 64, 64><<<grid, 64, 0, s>>>(          \
       (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
-      fdpq)
+      fdpq, ntiles, (int)pl.chunks, xcd_map)
 #define SEL(T, MODE, SIMP)                                                     \
   if (pl.tile == 64) RUN64(T, MODE, SIMP);                                     \
   else if (dbuf) RUN(T, MODE, SIMP, true);                                     \
