@@ -117,6 +117,37 @@ at::Tensor conv_fwd(at::Tensor x, at::Tensor w,
   return y;
 }
 
+// conv forward that also emits BN {sum,sumsq} partials from the MFMA
+// epilogue ([cdiv(M,128)][2][K] fp32) so the following BatchNorm skips its
+// full re-read of y.  Returns (y, partials); partials is EMPTY on the
+// paths that do not support fusion (fp32, C % 8 != 0) — caller falls back
+// to the standalone bn_stats pass.
+std::tuple<at::Tensor, at::Tensor> conv_fwd_stats(
+    at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias, int64_t sh,
+    int64_t sw, int64_t ph, int64_t pw, bool relu) {
+  check_cl(x, "x"); check_cl(w, "w");
+  TORCH_CHECK(x.scalar_type() == w.scalar_type());
+  Conv g = conv_geom(x, w, sh, sw, ph, pw);
+  auto y = empty_nhwc(g.N, g.K, g.P, g.Q, x.options());
+  const void* bptr = nullptr;
+  if (bias.has_value() && bias->defined()) bptr = bias->data_ptr();
+  int dt = dt_code(x);
+  const int epu = dt == 0 ? 4 : 8;
+  auto f32 = x.options().dtype(at::kFloat);
+  if (g.C % epu != 0 || dt == 0) {
+    auto yo = conv_fwd(x, w, bias, sh, sw, ph, pw, relu);
+    return {yo, at::empty({0}, f32)};
+  }
+  const int64_t M = (int64_t)g.N * g.P * g.Q;
+  const int64_t chunks = (M + 127) / 128;
+  auto partials = at::empty({chunks * 2 * g.K}, f32);
+  ddl_conv_fwd_stats(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr(),
+                     partials.data_ptr<float>(), g.N, g.H, g.W, g.C, g.K,
+                     g.R, g.S, g.P, g.Q, sh, sw, ph, pw, relu, dt,
+                     cur_stream());
+  return {y, partials};
+}
+
 // small-C (stem) variant returning the im2col buffer so backward reuses it
 std::tuple<at::Tensor, at::Tensor> conv_fwd_smallc(
     at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias, int64_t sh,
@@ -217,7 +248,8 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor rmean,
     at::Tensor rvar, double momentum, double eps, bool relu,
-    c10::optional<at::Tensor> residual) {
+    c10::optional<at::Tensor> residual,
+    c10::optional<at::Tensor> conv_partials) {
   check_cl(x, "x");
   int64_t C = x.size(1);
   int64_t M = x.numel() / C;
@@ -226,19 +258,32 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
               "running stats must stay fp32");
   TORCH_CHECK(weight.scalar_type() == x.scalar_type());
   auto f32 = x.options().dtype(at::kFloat);
-  auto sums = at::empty({2 * C}, f32);
-  int chunks = ddl_bn_stats_chunks(M, C);
-  auto partials = at::empty({(int64_t)chunks * 2 * C}, f32);
   auto mean = at::empty({C}, f32);
   auto rstd = at::empty({C}, f32);
   auto y = at::empty_like(x);
   auto s = cur_stream();
   int dt = dt_code(x);
-  ddl_bn_stats(x.data_ptr(), M, C, dt, sums.data_ptr<float>(),
-               partials.data_ptr<float>(), chunks, s);
-  ddl_bn_finalize(sums.data_ptr<float>(), C, M, momentum, eps,
-                  rmean.data_ptr<float>(), rvar.data_ptr<float>(),
-                  mean.data_ptr<float>(), rstd.data_ptr<float>(), 1, s);
+  if (conv_partials.has_value() && conv_partials->numel() > 0) {
+    // partials from the producing conv's epilogue: [chunks][2][C]
+    const int64_t pchunks = conv_partials->numel() / (2 * C);
+    TORCH_CHECK(pchunks * 2 * C == conv_partials->numel(),
+                "conv partials shape mismatch");
+    auto fold_ws = at::empty({(2 + 512) * C}, f32);
+    ddl_bn_reduce_finalize(conv_partials->data_ptr<float>(),
+                           fold_ws.data_ptr<float>(), (int)pchunks, C, M,
+                           momentum, eps, rmean.data_ptr<float>(),
+                           rvar.data_ptr<float>(), mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), 1, s);
+  } else {
+    auto sums = at::empty({2 * C}, f32);
+    int chunks = ddl_bn_stats_chunks(M, C);
+    auto partials = at::empty({(int64_t)chunks * 2 * C}, f32);
+    ddl_bn_stats(x.data_ptr(), M, C, dt, sums.data_ptr<float>(),
+                 partials.data_ptr<float>(), chunks, s);
+    ddl_bn_finalize(sums.data_ptr<float>(), C, M, momentum, eps,
+                    rmean.data_ptr<float>(), rvar.data_ptr<float>(),
+                    mean.data_ptr<float>(), rstd.data_ptr<float>(), 1, s);
+  }
   const void* res = residual.has_value() && residual->defined()
                         ? residual->data_ptr() : nullptr;
   ddl_bn_apply(x.data_ptr(), y.data_ptr(), res, mean.data_ptr<float>(),
@@ -443,11 +488,20 @@ void lars_norms(at::Tensor table, at::Tensor norms, int64_t dtype, double wd,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd);
+  m.def("conv_fwd_stats", &conv_fwd_stats);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);
   m.def("conv_fwd_smallc", &conv_fwd_smallc);
   m.def("conv_wgrad_smallc", &conv_wgrad_smallc);
   m.def("bn_relu_fwd", &bn_relu_fwd);
+  // 9-arg overload (no conv partials) for existing callers
+  m.def("bn_relu_fwd",
+        [](at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor rmean,
+           at::Tensor rvar, double momentum, double eps, bool relu,
+           c10::optional<at::Tensor> residual) {
+          return bn_relu_fwd(x, weight, bias, rmean, rvar, momentum, eps,
+                             relu, residual, c10::nullopt);
+        });
   m.def("bn_relu_infer", &bn_relu_infer);
   m.def("bn_relu_bwd", &bn_relu_bwd);
   m.def("maxpool_fwd", &maxpool_fwd);

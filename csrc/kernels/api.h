@@ -41,6 +41,13 @@ int ddl_bn_stats_chunks(int64_t M, int C);
 void ddl_bn_stats(const void* x, int64_t M, int C, int dtype, float* sums,
                   float* partials, int max_chunks, hipStream_t s);
 // finalize: mean/rstd from sums; optionally update running stats in-place.
+// fold conv-epilogue partials [chunks][2][C] -> sums[2][C] (deterministic,
+// two stage for large chunk counts) and finalize mean/rstd.
+void ddl_bn_reduce_finalize(const float* partials, float* fold_ws, int chunks,
+                            int C, int64_t M, float momentum, float eps,
+                            float* running_mean, float* running_var,
+                            float* mean_out, float* rstd_out,
+                            int update_running, hipStream_t s);
 void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
                      float eps, float* running_mean, float* running_var,
                      float* mean_out, float* rstd_out, int update_running,
@@ -95,6 +102,12 @@ void ddl_softmax_xent_bwd(const float* probs, const int64_t* target,
 
 // ---- implicit-GEMM conv (K1/K2/K3), NHWC, MFMA ----------------------------
 // x [N,H,W,C], w [K,R,S,C] (KRSC = channels_last), y [N,P,Q,K].
+// conv forward that also emits BN {sum,sumsq} partials [cdiv(M,128)][2][K]
+// from its epilogue (2-byte dtypes, vectorizable C; stats_ws may be null).
+void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
+                        void* y, float* stats_ws, int N, int H, int W, int C,
+                        int K, int R, int S, int P, int Q, int sh, int sw,
+                        int ph, int pw, int relu, int dtype, hipStream_t s);
 void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
                   int N, int H, int W, int C, int K, int R, int S, int P,
                   int Q, int sh, int sw, int ph, int pw, int relu, int dtype,

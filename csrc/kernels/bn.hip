@@ -113,6 +113,56 @@ __global__ void bn_reduce_partials_k(const float* __restrict__ partials,
   }
 }
 
+// fold partials[chunks][2][C] -> fold_ws[FOLD][2][C]: group g sums chunks
+// g, g+FOLD, ... (fixed order, deterministic).  The conv-epilogue partials
+// have one chunk PER 128-ROW M-TILE (tens of thousands for 56^2 layers);
+// bn_reduce_partials_k alone has only C/64 blocks of parallelism there.
+__global__ void bn_fold_partials_k(const float* __restrict__ partials,
+                                   float* __restrict__ fold_ws, int C,
+                                   int chunks, int fold) {
+  const int vg = threadIdx.x % CG;
+  const int trow = threadIdx.x / CG;
+  const int c0 = blockIdx.x * CG * 8 + vg * 8;
+  const int g = blockIdx.y;
+  if (c0 >= C) return;
+  __shared__ float red[2][ROWS][CG * 8];
+  float s0[8] = {0}, s1[8] = {0};
+  for (int i = g + trow * fold; i < chunks; i += fold * ROWS) {
+    const float* p0 = partials + ((int64_t)i * 2) * C + c0;
+    const float* p1 = p0 + C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      s0[j] += p0[j];
+      s1[j] += p1[j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][trow][vg * 8 + j] = s0[j];
+    red[1][trow][vg * 8 + j] = s1[j];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int step = ROWS / 2; step > 0; step >>= 1) {
+    if (trow < step) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        red[0][trow][vg * 8 + j] += red[0][trow + step][vg * 8 + j];
+        red[1][trow][vg * 8 + j] += red[1][trow + step][vg * 8 + j];
+      }
+    }
+    __syncthreads();
+  }
+  if (trow == 0) {
+    float* dst = fold_ws + ((int64_t)g * 2) * C;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      dst[c0 + j] = red[0][0][vg * 8 + j];
+      dst[C + c0 + j] = red[1][0][vg * 8 + j];
+    }
+  }
+}
+
 __global__ void bn_finalize_k(const float* __restrict__ sums, int C,
                               float inv_M, float unbias, float momentum,
                               float eps, float* running_mean,
@@ -346,6 +396,29 @@ void ddl_bn_finalize(const float* sums, int C, int64_t M, float momentum,
                      float eps, float* running_mean, float* running_var,
                      float* mean_out, float* rstd_out, int update_running,
                      hipStream_t s) {
+  float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
+  bn_finalize_k<<<cdiv(C, 256), 256, 0, s>>>(
+      sums, C, 1.f / (float)M, unbias, momentum, eps, running_mean,
+      running_var, mean_out, rstd_out, update_running);
+}
+
+void ddl_bn_reduce_finalize(const float* partials, float* fold_ws, int chunks,
+                            int C, int64_t M, float momentum, float eps,
+                            float* running_mean, float* running_var,
+                            float* mean_out, float* rstd_out,
+                            int update_running, hipStream_t s) {
+  // sums land in the first 2*C of fold_ws after the final reduce
+  float* sums = fold_ws;
+  if (chunks > 512) {
+    const int fold = 256;
+    float* stage = fold_ws + 2 * (int64_t)C;  // [fold][2][C]
+    dim3 g(cdiv(C, 64), fold);
+    bn_fold_partials_k<<<g, 256, 0, s>>>(partials, stage, C, chunks, fold);
+    bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(stage, sums, C, fold);
+  } else {
+    bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(partials, sums, C,
+                                                     chunks);
+  }
   float unbias = M > 1 ? (float)M / (float)(M - 1) : 1.f;
   bn_finalize_k<<<cdiv(C, 256), 256, 0, s>>>(
       sums, C, 1.f / (float)M, unbias, momentum, eps, running_mean,

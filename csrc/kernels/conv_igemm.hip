@@ -43,11 +43,16 @@ __device__ __forceinline__ void glds16(const void* gsrc, void* lds_base) {
                                    0, 0);
 }
 
+// stats_ws (FWD, 2-byte dtypes only): per-m-tile BN partials
+// [cdiv(M,BM)][2][Nc] {sum, sumsq} of the fp32 accumulators — the following
+// BatchNorm then skips its full re-read of y (bn_stats_k) and just folds
+// these (stats of the pre-bf16-rounding values; the difference is bf16
+// rounding noise, orders below BN's eps).
 template <typename T, int MODE, bool DBUF>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
-    int Nc, int relu, DgradPhase gp) {
+    int Nc, int relu, DgradPhase gp, float* __restrict__ stats_ws) {
   __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
   char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
   char* const lB[2] = {smem + TILE_BYTES,
@@ -381,6 +386,8 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   // 2-byte dtypes: repack through LDS, store coalesced 16 B rows
   __syncthreads();
   T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
+  const bool emit_stats = (MODE == FWD) && stats_ws != nullptr;
+  float cs[4] = {}, cq[4] = {};  // per-nr column {sum, sumsq} (BN fusion)
 #pragma unroll
   for (int nr = 0; nr < 4; ++nr) {
     const int n_loc = wc + nr * 16 + en;
@@ -392,6 +399,10 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       for (int reg = 0; reg < 4; ++reg) {
         float f = acc[mr][nr][reg] + bv;
         if (relu) f = fmaxf(f, 0.f);
+        if (emit_stats) {
+          cs[nr] += f;
+          cq[nr] += f * f;
+        }
         etile[(wr + mr * 16 + em + reg) * BN + n_loc] = from_f32<T>(f);
       }
     }
@@ -408,6 +419,38 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       *reinterpret_cast<uint4*>(out + out_row_elem(m) + tile_n + ub) =
           *reinterpret_cast<const uint4*>(etile + row * BN + ub);
     }
+  }
+  if (!emit_stats) return;
+  // fold the per-thread column sums: lanes l, l+16, l+32, l+48 share a
+  // column (em groups) -> xor-shuffle; then the two m-half waves combine
+  // through LDS (reused as scratch after the store loop).
+#pragma unroll
+  for (int nr = 0; nr < 4; ++nr) {
+#pragma unroll
+    for (int off = 16; off <= 32; off <<= 1) {
+      cs[nr] += __shfl_xor(cs[nr], off, 64);
+      cq[nr] += __shfl_xor(cq[nr], off, 64);
+    }
+  }
+  __syncthreads();
+  float* sred = reinterpret_cast<float*>(smem);  // [4 wid][4 nr][16 en][2]
+  if (lane < 16) {
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      sred[((wid * 4 + nr) * 16 + lane) * 2 + 0] = cs[nr];
+      sred[((wid * 4 + nr) * 16 + lane) * 2 + 1] = cq[nr];
+    }
+  }
+  __syncthreads();
+  // col = wcg*64 + nr*16 + en is covered by waves {wcg, wcg+2}
+  const int col = threadIdx.x;
+  if (col < BN && tile_n + col < Nc) {
+    const int wcg = col >> 6, nr = (col >> 4) & 3, e = col & 15;
+    const int i0 = ((wcg * 4 + nr) * 16 + e) * 2;
+    const int i1 = (((wcg + 2) * 4 + nr) * 16 + e) * 2;
+    float* dst = stats_ws + (int64_t)(bid / nt_n) * 2 * Nc + tile_n + col;
+    dst[0] = sred[i0] + sred[i1];
+    dst[Nc] = sred[i0 + 1] + sred[i1 + 1];
   }
 }
 
@@ -426,16 +469,28 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
                   int N, int H, int W, int C, int K, int R, int S, int P,
                   int Q, int sh, int sw, int ph, int pw, int relu, int dtype,
                   hipStream_t s) {
+  ddl_conv_fwd_stats(x, w, bias, y, nullptr, N, H, W, C, K, R, S, P, Q, sh,
+                     sw, ph, pw, relu, dtype, s);
+}
+
+// stats_ws != nullptr (2-byte dtypes, vectorizable C): also emit the
+// [cdiv(M,128)][2][K] fp32 BN {sum,sumsq} partials from the epilogue.
+void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
+                        void* y, float* stats_ws, int N, int H, int W, int C,
+                        int K, int R, int S, int P, int Q, int sh, int sw,
+                        int ph, int pw, int relu, int dtype, hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int64_t M = (int64_t)N * P * Q;
   const int Nc = K;
   const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
   const bool dbuf = conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
+  if (flat || dtype == 0) stats_ws = nullptr;  // unsupported paths
   DgradPhase gp{};  // unused by FWD modes
 #define RUN(T, MODE, DB)                                                       \
   conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
-      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp)
+      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp,    \
+      stats_ws)
 #define SEL(T, MODE)                                                           \
   if (dbuf) RUN(T, MODE, true); else RUN(T, MODE, false)
   if (dtype == 0) { if (flat) { SEL(float, FWD_FLAT); } else { SEL(float, FWD); } }
@@ -457,7 +512,7 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
     DgradPhase gp{};
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp)
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
     if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
     else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
     else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
@@ -481,7 +536,7 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
       const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD_S2, DB><<<(unsigned)grid, 256, 0, s>>>(                \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp)
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
       if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
       else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
       else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }

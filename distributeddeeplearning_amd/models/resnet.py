@@ -24,11 +24,13 @@ from ..ops.modules import (
 
 
 def conv3x3(in_planes, out_planes, stride=1):
-    return Conv2d(in_planes, out_planes, 3, stride=stride, padding=1, bias=False)
+    return Conv2d(in_planes, out_planes, 3, stride=stride, padding=1,
+                  bias=False, emit_bn_stats=True)
 
 
 def conv1x1(in_planes, out_planes, stride=1):
-    return Conv2d(in_planes, out_planes, 1, stride=stride, bias=False)
+    return Conv2d(in_planes, out_planes, 1, stride=stride, bias=False,
+                  emit_bn_stats=True)
 
 
 class BasicBlock(nn.Module):
@@ -83,7 +85,8 @@ class ResNet(nn.Module):
     ):
         super().__init__()
         self.inplanes = 64
-        self.conv1 = Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
+        self.conv1 = Conv2d(3, 64, 7, stride=2, padding=3, bias=False,
+                            emit_bn_stats=True)
         self.bn1 = BatchNorm2d(64)
         self.maxpool = MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, layers[0])

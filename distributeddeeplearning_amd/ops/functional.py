@@ -38,12 +38,13 @@ def to_channels_last(t: torch.Tensor) -> torch.Tensor:
 
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, stride, padding, relu):
+    def forward(ctx, x, weight, bias, stride, padding, relu, emit_stats=False):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         weight = to_channels_last(weight)
         epu = 4 if x.dtype == torch.float32 else 8
         ctx.small_c = x.shape[1] % epu != 0
+        partials = None
         if ctx.small_c:
             # stem path: keep the im2col buffer for wgrad reuse
             y, col = ext.conv_fwd_smallc(
@@ -51,8 +52,18 @@ class _Conv2dFn(torch.autograd.Function):
                 stride[0], stride[1], padding[0], padding[1], relu)
             saved = [x, weight, col]
         else:
-            y = ext.conv_fwd(x, weight, bias if bias is not None else None,
-                             stride[0], stride[1], padding[0], padding[1], relu)
+            if emit_stats:
+                # K1+K4 fusion: the MFMA epilogue also emits the following
+                # BatchNorm's {sum, sumsq} partials, skipping bn_stats_k's
+                # full re-read of y (empty partials = unsupported path)
+                y, partials = ext.conv_fwd_stats(
+                    x, weight, bias if bias is not None else None,
+                    stride[0], stride[1], padding[0], padding[1], relu)
+            else:
+                y = ext.conv_fwd(x, weight,
+                                 bias if bias is not None else None,
+                                 stride[0], stride[1], padding[0],
+                                 padding[1], relu)
             saved = [x, weight]
         if relu:
             saved.append(y)
@@ -61,10 +72,17 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.padding = padding
         ctx.has_bias = bias is not None
         ctx.relu = relu
+        ctx.emit_stats = emit_stats
+        if emit_stats:
+            if partials is None:
+                partials = torch.empty(0, dtype=torch.float32,
+                                       device=x.device)
+            ctx.mark_non_differentiable(partials)
+            return y, partials
         return y
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, *unused_dpartials):
         ext = load_kernels_ext()
         saved = list(ctx.saved_tensors)
         if ctx.relu:
@@ -96,15 +114,27 @@ class _Conv2dFn(torch.autograd.Function):
                                     ctx.padding[0], ctx.padding[1])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.sum(dim=(0, 2, 3))
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
 
 
 def conv2d(x, weight, bias=None, stride=1, padding=0, relu=False):
     stride, padding = _pair(stride), _pair(padding)
     if x.is_cuda:
-        return _Conv2dFn.apply(x, weight, bias, stride, padding, relu)
+        return _Conv2dFn.apply(x, weight, bias, stride, padding, relu, False)
     y = F.conv2d(x, weight, bias, stride, padding)
     return F.relu(y) if relu else y
+
+
+def conv2d_stats(x, weight, bias=None, stride=1, padding=0):
+    """conv2d that also returns the following BatchNorm's {sum, sumsq}
+    partials computed in the conv epilogue (None when unsupported: CPU,
+    fp32, or the C%8!=0 stem)."""
+    stride, padding = _pair(stride), _pair(padding)
+    if x.is_cuda:
+        y, partials = _Conv2dFn.apply(x, weight, bias, stride, padding,
+                                      False, True)
+        return y, (partials if partials.numel() > 0 else None)
+    return F.conv2d(x, weight, bias, stride, padding), None
 
 
 # --------------------------------------------------------------------------
@@ -114,7 +144,7 @@ def conv2d(x, weight, bias=None, stride=1, padding=0, relu=False):
 class _BnAddReluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                momentum, eps, relu, residual):
+                momentum, eps, relu, residual, conv_partials=None):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         if residual is not None:
@@ -123,6 +153,7 @@ class _BnAddReluFn(torch.autograd.Function):
             x, weight, bias, running_mean, running_var,
             float(momentum), float(eps), relu,
             residual if residual is not None else None,
+            conv_partials,
         )
         ctx.save_for_backward(x, weight, bias, mean, rstd, y)
         ctx.relu = relu
@@ -137,11 +168,12 @@ class _BnAddReluFn(torch.autograd.Function):
         dx, dw, db, dres = ext.bn_relu_bwd(
             dy, x, weight, bias, mean, rstd, y, ctx.relu, ctx.has_residual)
         return (dx, dw, db, None, None, None, None, None,
-                dres if ctx.has_residual else None)
+                dres if ctx.has_residual else None, None)
 
 
 def bn_add_relu(x, weight, bias, running_mean, running_var, training,
-                momentum=0.1, eps=1e-5, relu=True, residual=None):
+                momentum=0.1, eps=1e-5, relu=True, residual=None,
+                conv_partials=None):
     """y = relu?(bn(x) + residual?) — single fused kernel each direction.
 
     The reference issues separate cuDNN BN and ReLU (and a separate add for
@@ -150,7 +182,8 @@ def bn_add_relu(x, weight, bias, running_mean, running_var, training,
     if x.is_cuda:
         if training:
             return _BnAddReluFn.apply(x, weight, bias, running_mean,
-                                      running_var, momentum, eps, relu, residual)
+                                      running_var, momentum, eps, relu,
+                                      residual, conv_partials)
         ext = load_kernels_ext()
         return ext.bn_relu_infer(to_channels_last(x), weight, bias,
                                  running_mean, running_var, float(eps), relu,

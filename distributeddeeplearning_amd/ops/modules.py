@@ -17,8 +17,11 @@ from . import functional as DF
 
 class Conv2d(nn.Module):
     def __init__(self, in_channels, out_channels, kernel_size, stride=1,
-                 padding=0, bias=False):
+                 padding=0, bias=False, emit_bn_stats=False):
         super().__init__()
+        # emit_bn_stats: this conv feeds a BatchNorm — the MFMA epilogue
+        # also emits the BN {sum,sumsq} partials so bn skips its stats pass
+        self.emit_bn_stats = emit_bn_stats
         k = (kernel_size, kernel_size) if isinstance(kernel_size, int) else kernel_size
         self.in_channels = in_channels
         self.out_channels = out_channels
@@ -36,6 +39,12 @@ class Conv2d(nn.Module):
         nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
 
     def forward(self, x):
+        if self.emit_bn_stats and self.training and x.is_cuda:
+            y, partials = DF.conv2d_stats(x, self.weight, self.bias,
+                                          self.stride, self.padding)
+            if partials is not None:
+                y._ddl_bn_partials = partials
+            return y
         return DF.conv2d(x, self.weight, self.bias, self.stride, self.padding)
 
     def extra_repr(self):
@@ -67,9 +76,13 @@ class BatchNorm2d(nn.Module):
     def forward(self, x, relu: bool = False, residual: Optional[torch.Tensor] = None):
         if self.training:
             self.num_batches_tracked += 1
+        partials = getattr(x, "_ddl_bn_partials", None)
+        if partials is not None:
+            del x._ddl_bn_partials
         return DF.bn_add_relu(
             x, self.weight, self.bias, self.running_mean, self.running_var,
-            self.training, self.momentum, self.eps, relu, residual)
+            self.training, self.momentum, self.eps, relu, residual,
+            conv_partials=partials)
 
     def extra_repr(self):
         return f"{self.num_features}, eps={self.eps}, momentum={self.momentum}"

@@ -299,3 +299,57 @@ def test_bn_relu_no_residual_bwd_recomputed_mask():
     assert relerr(dw, wgt.grad) < 0.03
     assert relerr(db, b.grad) < 0.03
     assert relerr(dx, x.grad) < 0.06
+
+
+def test_conv_bn_stats_fusion():
+    """K1+K4 fusion: BN mean/rstd folded from the conv epilogue's partials
+    match the standalone bn_stats pass over the conv output."""
+    ext = _ext()
+    x = cl(mk((8, 64, 28, 28), seed=41).to(DEV))
+    w = cl(mk((128, 64, 3, 3), seed=42).to(DEV) * 0.1)
+    y1, partials = ext.conv_fwd_stats(x, w, None, 1, 1, 1, 1, False)
+    y0 = ext.conv_fwd(x, w, None, 1, 1, 1, 1, False)
+    assert torch.equal(y1, y0), "fused-stats conv changed y"
+    assert partials.numel() > 0
+    C = 128
+    M = y1.numel() // C
+    assert partials.numel() % (2 * C) == 0
+    wgt = torch.ones(C, dtype=torch.bfloat16, device=DEV)
+    b = torch.zeros(C, dtype=torch.bfloat16, device=DEV)
+    rm0, rv0 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+    rm1, rv1 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
+    yb0, mean0, rstd0 = ext.bn_relu_fwd(y0, wgt, b, rm0, rv0, 0.1, 1e-5,
+                                        False, None)
+    yb1, mean1, rstd1 = ext.bn_relu_fwd(y1, wgt, b, rm1, rv1, 0.1, 1e-5,
+                                        False, None, partials)
+    # partials are pre-bf16-rounding sums: equal up to rounding noise
+    assert relerr(mean1, mean0) < 5e-3
+    assert relerr(rstd1, rstd0) < 5e-3
+    assert relerr(yb1, yb0) < 5e-3
+    assert relerr(rv1, rv0) < 5e-3
+
+
+def test_conv_bn_fusion_module_training_step():
+    """Module-level wiring: resnet conv(emit_bn_stats) -> bn trains and
+    matches the unfused configuration closely."""
+    from distributeddeeplearning_amd.ops.modules import BatchNorm2d, Conv2d
+    torch.manual_seed(7)
+    conv_f = Conv2d(64, 128, 3, padding=1, emit_bn_stats=True).to(DEV)
+    bn_f = BatchNorm2d(128).to(DEV)
+    conv_u = Conv2d(64, 128, 3, padding=1, emit_bn_stats=False).to(DEV)
+    bn_u = BatchNorm2d(128).to(DEV)
+    conv_u.load_state_dict(conv_f.state_dict())
+    bn_u.load_state_dict(bn_f.state_dict())
+    from distributeddeeplearning_amd.ops.modules import cast_model_
+    for m in (conv_f, bn_f, conv_u, bn_u):
+        cast_model_(m, torch.bfloat16)
+        m.train()
+    x = cl(mk((8, 64, 14, 14), seed=43).to(DEV)).requires_grad_(True)
+    x2 = x.detach().clone().requires_grad_(True)
+    yf = bn_f(conv_f(x), relu=True)
+    yu = bn_u(conv_u(x2), relu=True)
+    assert relerr(yf, yu) < 5e-3
+    gf = torch.autograd.grad(yf.float().sum(), [x, conv_f.weight])
+    gu = torch.autograd.grad(yu.float().sum(), [x2, conv_u.weight])
+    assert relerr(gf[0], gu[0]) < 2e-2
+    assert relerr(gf[1], gu[1]) < 2e-2
