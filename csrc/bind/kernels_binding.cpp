@@ -138,8 +138,8 @@ std::tuple<at::Tensor, at::Tensor> conv_fwd_stats(
     auto yo = conv_fwd(x, w, bias, sh, sw, ph, pw, relu);
     return {yo, at::empty({0}, f32)};
   }
-  const int64_t M = (int64_t)g.N * g.P * g.Q;
-  const int64_t chunks = (M + 127) / 128;
+  const int64_t chunks = ddl_conv_fwd_stats_chunks(g.N, g.P, g.Q, g.C, g.K,
+                                                   g.R, g.S, dt);
   auto partials = at::empty({chunks * 2 * g.K}, f32);
   ddl_conv_fwd_stats(x.data_ptr(), w.data_ptr(), bptr, y.data_ptr(),
                      partials.data_ptr<float>(), g.N, g.H, g.W, g.C, g.K,

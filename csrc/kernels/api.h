@@ -108,6 +108,8 @@ void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
                         void* y, float* stats_ws, int N, int H, int W, int C,
                         int K, int R, int S, int P, int Q, int sh, int sw,
                         int ph, int pw, int relu, int dtype, hipStream_t s);
+int64_t ddl_conv_fwd_stats_chunks(int N, int P, int Q, int C, int K, int R,
+                                  int S, int dtype);
 void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
                   int N, int H, int W, int C, int K, int R, int S, int P,
                   int Q, int sh, int sw, int ph, int pw, int relu, int dtype,

@@ -48,34 +48,44 @@ __device__ __forceinline__ void glds16(const void* gsrc, void* lds_base) {
 // BatchNorm then skips its full re-read of y (bn_stats_k) and just folds
 // these (stats of the pre-bf16-rounding values; the difference is bf16
 // rounding noise, orders below BN's eps).
-template <typename T, int MODE, bool DBUF>
+// TBM/TBN: block tile.  (128,128) = the standard 2x2-wave tile; (256,64)
+// puts the 4 waves in a 4x1 m-column over a 64-wide n-tile for Nc <= 64
+// outputs (ResNet layer1 / the im2col'd stem: a 128-wide n-tile there runs
+// 50% dead MFMA work).  SBUF only (A tile is 32 KiB at TBM=256).
+template <typename T, int MODE, bool DBUF, int TBM = 128, int TBN = 128>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
     int Nc, int relu, DgradPhase gp, float* __restrict__ stats_ws) {
-  __shared__ __align__(16) char smem[DBUF ? 4 * TILE_BYTES : 2 * TILE_BYTES];
-  char* const lA[2] = {smem, DBUF ? smem + 2 * TILE_BYTES : smem};
-  char* const lB[2] = {smem + TILE_BYTES,
-                       DBUF ? smem + 3 * TILE_BYTES : smem + TILE_BYTES};
+  constexpr int AROWS = TBM / 32;     // staged A rows per thread
+  constexpr int BROWS = TBN / 32;     // staged B rows per thread
+  constexpr int ATB = TBM * 128;      // A tile bytes (128 B rows)
+  constexpr int BTB = TBN * 128;      // B tile bytes
+  static_assert(!(DBUF && TBM != 128), "DBUF only for the 128^2 tile");
+  __shared__ __align__(16) char smem[DBUF ? 2 * (ATB + BTB) : ATB + BTB];
+  char* const lA[2] = {smem, DBUF ? smem + ATB + BTB : smem};
+  char* const lB[2] = {smem + ATB,
+                       DBUF ? smem + 2 * ATB + BTB : smem + ATB};
 
-  const int nt_n = (Nc + BN - 1) / BN;
+  const int nt_n = (Nc + TBN - 1) / TBN;
   const unsigned bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  const int64_t tile_m = (int64_t)(bid / nt_n) * BM;
-  const int tile_n = (int)(bid % nt_n) * BN;
+  const int64_t tile_m = (int64_t)(bid / nt_n) * TBM;
+  const int tile_n = (int)(bid % nt_n) * TBN;
 
   const int lane = threadIdx.x & 63;
   const int wid = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
-  const int wr = ((threadIdx.x >> 6) >> 1) * 64;
-  const int wc = ((threadIdx.x >> 6) & 1) * 64;
+  const int wr = (TBN == 64) ? (threadIdx.x >> 6) * 64
+                             : ((threadIdx.x >> 6) >> 1) * 64;
+  const int wc = (TBN == 64) ? 0 : ((threadIdx.x >> 6) & 1) * 64;
   const int trow = threadIdx.x >> 3;   // this thread's staging row base
   const int tcol = threadIdx.x & 7;    // 16B unit within the row
 
   f32x4_t acc[4][4] = {};
 
-  // per-thread A-row decode (4 staged rows: trow + 32*i)
-  int a_n[4], a_i1[4], a_i2[4];
+  // per-thread A-row decode (AROWS staged rows: trow + 32*i)
+  int a_n[AROWS], a_i1[AROWS], a_i2[AROWS];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < AROWS; ++i) {
     int64_t m = tile_m + trow + 32 * i;
     if (m < M) {
       if (MODE == DGRAD_S2) {
@@ -112,8 +122,8 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   const int total_tiles = rs_total * nct;
 
   int cur_rs = -1;
-  const T* a_ptr[4];
-  const T* b_ptr[4];
+  const T* a_ptr[AROWS];
+  const T* b_ptr[BROWS];
   const T* zpad = reinterpret_cast<const T*>(ddl_zero_page);
 
   auto update_rs_ptrs = [&](int rs) {
@@ -127,7 +137,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
     cur_rs = rs;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < AROWS; ++i) {
       a_ptr[i] = nullptr;
       if (a_n[i] >= 0) {
         if (MODE == FWD) {
@@ -158,7 +168,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       }
     }
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < BROWS; ++i) {
       const int row_n = tile_n + trow + 32 * i;
       b_ptr[i] = nullptr;
       if (row_n < Nc) {
@@ -183,12 +193,12 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     const int cel = c0 + scu * EPU;
     const bool cok = cel + EPU <= Cred;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < AROWS; ++i) {
       const T* src = (a_ptr[i] != nullptr && cok) ? a_ptr[i] + cel : zpad;
       glds16(src, la + wid * 1024 + i * 4096);
     }
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < BROWS; ++i) {
       const T* src = (b_ptr[i] != nullptr && cok) ? b_ptr[i] + cel : zpad;
       glds16(src, lb + wid * 1024 + i * 4096);
     }
@@ -385,7 +395,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   }
   // 2-byte dtypes: repack through LDS, store coalesced 16 B rows
   __syncthreads();
-  T* etile = reinterpret_cast<T*>(smem);  // [BM][BN] elements (32 KiB)
+  T* etile = reinterpret_cast<T*>(smem);  // [TBM][TBN] elements (32 KiB)
   const bool emit_stats = (MODE == FWD) && stats_ws != nullptr;
   float cs[4] = {}, cq[4] = {};  // per-nr column {sum, sumsq} (BN fusion)
 #pragma unroll
@@ -403,21 +413,22 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
           cs[nr] += f;
           cq[nr] += f * f;
         }
-        etile[(wr + mr * 16 + em + reg) * BN + n_loc] = from_f32<T>(f);
+        etile[(wr + mr * 16 + em + reg) * TBN + n_loc] = from_f32<T>(f);
       }
     }
   }
   __syncthreads();
   const int nvalid = Nc - tile_n;
+  constexpr int UPR = TBN / 8;  // 16 B units per etile row
 #pragma unroll
-  for (int it = 0; it < 8; ++it) {
+  for (int it = 0; it < TBM * TBN / 8 / 256; ++it) {
     const int unit = threadIdx.x + it * 256;
-    const int row = unit >> 4;
-    const int ub = (unit & 15) * 8;
+    const int row = unit / UPR;
+    const int ub = (unit % UPR) * 8;
     const int64_t m = tile_m + row;
     if (m < M && ub < nvalid) {
       *reinterpret_cast<uint4*>(out + out_row_elem(m) + tile_n + ub) =
-          *reinterpret_cast<const uint4*>(etile + row * BN + ub);
+          *reinterpret_cast<const uint4*>(etile + row * TBN + ub);
     }
   }
   if (!emit_stats) return;
@@ -442,16 +453,37 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     }
   }
   __syncthreads();
-  // col = wcg*64 + nr*16 + en is covered by waves {wcg, wcg+2}
   const int col = threadIdx.x;
-  if (col < BN && tile_n + col < Nc) {
-    const int wcg = col >> 6, nr = (col >> 4) & 3, e = col & 15;
-    const int i0 = ((wcg * 4 + nr) * 16 + e) * 2;
-    const int i1 = (((wcg + 2) * 4 + nr) * 16 + e) * 2;
+  if (col < TBN && tile_n + col < Nc) {
+    const int nr = (col >> 4) & 3, e = col & 15;
     float* dst = stats_ws + (int64_t)(bid / nt_n) * 2 * Nc + tile_n + col;
-    dst[0] = sred[i0] + sred[i1];
-    dst[Nc] = sred[i0 + 1] + sred[i1 + 1];
+    if (TBN == 64) {
+      // 4x1 wave column: every wave covers every column
+      float s = 0.f, q = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        s += sred[((w * 4 + nr) * 16 + e) * 2];
+        q += sred[((w * 4 + nr) * 16 + e) * 2 + 1];
+      }
+      dst[0] = s;
+      dst[Nc] = q;
+    } else {
+      // col = wcg*64 + nr*16 + en is covered by waves {wcg, wcg+2}
+      const int wcg = col >> 6;
+      const int i0 = ((wcg * 4 + nr) * 16 + e) * 2;
+      const int i1 = (((wcg + 2) * 4 + nr) * 16 + e) * 2;
+      dst[0] = sred[i0] + sred[i1];
+      dst[Nc] = sred[i0 + 1] + sred[i1 + 1];
+    }
   }
+}
+
+// (256,64) n-narrow tile: Nc <= 64 outputs with enough m-tiles to fill
+// the chip (ResNet layer1 64-out convs, their dgrads, the im2col'd stem).
+bool conv_use_n64(int Nc, int64_t M) {
+  const char* env = getenv("DDL_CONV_N64");
+  if (env && env[0] == '0') return false;
+  return Nc <= 64 && M / 256 >= 512;
 }
 
 bool conv_use_dbuf(int cred, int rs_total) {
@@ -482,22 +514,42 @@ void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int64_t M = (int64_t)N * P * Q;
   const int Nc = K;
-  const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
-  const bool dbuf = conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
+  const bool n64 = !flat && conv_use_n64(Nc, M);
+  const int64_t grid =
+      n64 ? cdiv(M, 256) * cdiv(Nc, 64) : cdiv(M, BM) * cdiv(Nc, BN);
+  const bool dbuf =
+      !n64 && conv_use_dbuf(flat ? R * S * C : C, flat ? 1 : R * S);
   if (flat || dtype == 0) stats_ws = nullptr;  // unsupported paths
   DgradPhase gp{};  // unused by FWD modes
 #define RUN(T, MODE, DB)                                                       \
   conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
       (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp,    \
       stats_ws)
+#define RUN64(T, MODE)                                                         \
+  conv_igemm_k<T, MODE, false, 256, 64><<<(unsigned)grid, 256, 0, s>>>(        \
+      (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp,    \
+      stats_ws)
 #define SEL(T, MODE)                                                           \
-  if (dbuf) RUN(T, MODE, true); else RUN(T, MODE, false)
+  if (n64) RUN64(T, MODE);                                                     \
+  else if (dbuf) RUN(T, MODE, true);                                           \
+  else RUN(T, MODE, false)
   if (dtype == 0) { if (flat) { SEL(float, FWD_FLAT); } else { SEL(float, FWD); } }
   else if (dtype == 1) { if (flat) { SEL(bf16, FWD_FLAT); } else { SEL(bf16, FWD); } }
   else if (dtype == 2) { if (flat) { SEL(f16, FWD_FLAT); } else { SEL(f16, FWD); } }
 #undef SEL
+#undef RUN64
 #undef RUN
+}
+
+// chunk count of the BN partials a fused conv forward will emit
+// (must mirror the tile selection above)
+int64_t ddl_conv_fwd_stats_chunks(int N, int P, int Q, int C, int K, int R,
+                                  int S, int dtype) {
+  const int64_t M = (int64_t)N * P * Q;
+  const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
+  const bool n64 = !flat && conv_use_n64(K, M);
+  return n64 ? cdiv(M, 256) : cdiv(M, BM);
 }
 
 void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
@@ -508,14 +560,20 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
   const bool dbuf = conv_use_dbuf(K, R * S);
   if (sh == 1 && sw == 1) {
     const int64_t M = (int64_t)N * H * W;
-    const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+    const bool n64 = conv_use_n64(Nc, M);
+    const int64_t grid =
+        n64 ? cdiv(M, 256) * cdiv(Nc, 64) : cdiv(M, BM) * cdiv(Nc, BN);
     DgradPhase gp{};
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
       (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
-    if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
-    else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
-    else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
+#define RUN64(T)                                                               \
+  conv_igemm_k<T, DGRAD, false, 256, 64><<<(unsigned)grid, 256, 0, s>>>(       \
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
+    if (dtype == 0) { if (n64) RUN64(float); else if (dbuf) RUN(float, true); else RUN(float, false); }
+    else if (dtype == 1) { if (n64) RUN64(bf16); else if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
+    else if (dtype == 2) { if (n64) RUN64(f16); else if (dbuf) RUN(f16, true); else RUN(f16, false); }
+#undef RUN64
 #undef RUN
     return;
   }
