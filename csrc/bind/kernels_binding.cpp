@@ -148,8 +148,10 @@ std::tuple<at::Tensor, at::Tensor> conv_fwd_stats(
   return {y, partials};
 }
 
-// small-C (stem) variant returning the im2col buffer so backward reuses it
-std::tuple<at::Tensor, at::Tensor> conv_fwd_smallc(
+// small-C (stem) variant returning the im2col buffer so backward reuses
+// it, plus the BN stats partials from the (flat-GEMM) epilogue — the stem
+// feeds bn1 like every other conv (empty when unsupported: fp32).
+std::tuple<at::Tensor, at::Tensor, at::Tensor> conv_fwd_smallc(
     at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias, int64_t sh,
     int64_t sw, int64_t ph, int64_t pw, bool relu) {
   check_cl(x, "x"); check_cl(w, "w");
@@ -163,10 +165,21 @@ std::tuple<at::Tensor, at::Tensor> conv_fwd_smallc(
   auto y = empty_nhwc(g.N, g.K, g.P, g.Q, x.options());
   const void* bptr = (bias.has_value() && bias->defined())
                          ? bias->data_ptr() : nullptr;
-  ddl_conv_fwd(col.data_ptr(), wp.data_ptr(), bptr, y.data_ptr(), g.N, g.P,
-               g.Q, ld, g.K, 1, 1, g.P, g.Q, 1, 1, 0, 0, relu, dt,
-               cur_stream());
-  return {y, col};
+  auto f32 = x.options().dtype(at::kFloat);
+  at::Tensor partials;
+  float* pptr = nullptr;
+  if (dt != 0) {
+    const int64_t chunks = ddl_conv_fwd_stats_chunks(g.N, g.P, g.Q, ld, g.K,
+                                                     1, 1, dt);
+    partials = at::empty({chunks * 2 * g.K}, f32);
+    pptr = partials.data_ptr<float>();
+  } else {
+    partials = at::empty({0}, f32);
+  }
+  ddl_conv_fwd_stats(col.data_ptr(), wp.data_ptr(), bptr, y.data_ptr(), pptr,
+                     g.N, g.P, g.Q, ld, g.K, 1, 1, g.P, g.Q, 1, 1, 0, 0,
+                     relu, dt, cur_stream());
+  return {y, col, partials};
 }
 
 at::Tensor conv_wgrad_smallc(at::Tensor dy, at::Tensor col,

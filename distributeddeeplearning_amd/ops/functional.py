@@ -47,9 +47,11 @@ class _Conv2dFn(torch.autograd.Function):
         partials = None
         if ctx.small_c:
             # stem path: keep the im2col buffer for wgrad reuse
-            y, col = ext.conv_fwd_smallc(
+            y, col, partials = ext.conv_fwd_smallc(
                 x, weight, bias if bias is not None else None,
                 stride[0], stride[1], padding[0], padding[1], relu)
+            if not emit_stats:
+                partials = None
             saved = [x, weight, col]
         else:
             if emit_stats:

@@ -68,6 +68,6 @@ def test_conv_shape_fwd_bwd(c, h, k, r, s, p):
         assert relerr(dx, x.grad) < 0.03, f"dgrad {c},{h},{k},{r},{s},{p}"
         dw = ext.conv_wgrad(dyg, xg, list(wt.shape), s, s, p, p)
     else:
-        _, col = ext.conv_fwd_smallc(xg, wg, None, s, s, p, p, False)
+        _, col, _p = ext.conv_fwd_smallc(xg, wg, None, s, s, p, p, False)
         dw = ext.conv_wgrad_smallc(dyg, col, list(wt.shape), list(x.shape))
     assert relerr(dw, wt.grad) < 0.03, f"wgrad {c},{h},{k},{r},{s},{p}"
