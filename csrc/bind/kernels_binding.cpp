@@ -206,7 +206,8 @@ at::Tensor conv_wgrad_smallc(at::Tensor dy, at::Tensor col,
 
 at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
                       std::vector<int64_t> x_shape, int64_t sh, int64_t sw,
-                      int64_t ph, int64_t pw) {
+                      int64_t ph, int64_t pw,
+                      c10::optional<at::Tensor> accum = c10::nullopt) {
   check_cl(dy, "dy"); check_cl(w, "w");
   int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
   int K = w.size(0), R = w.size(2), S = w.size(3);
@@ -218,8 +219,17 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
   ddl_transpose_krsc_crsk(w.data_ptr(), w_t.data_ptr(), K, R * S, C,
                           dt_code(w), cur_stream());
   auto dx = empty_nhwc(N, C, H, W, dy.options());
-  ddl_conv_dgrad(dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(), N, H, W, C, K,
-                 R, S, P, Q, sh, sw, ph, pw, dt_code(dy), cur_stream());
+  const void* aptr = nullptr;
+  if (accum.has_value() && accum->defined()) {
+    check_cl(*accum, "accum");
+    TORCH_CHECK(accum->scalar_type() == dy.scalar_type() &&
+                    accum->numel() == dx.numel(),
+                "dgrad accum shape/dtype mismatch");
+    aptr = accum->data_ptr();
+  }
+  ddl_conv_dgrad_accum(dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(), aptr,
+                       N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw,
+                       dt_code(dy), cur_stream());
   return dx;
 }
 
@@ -503,6 +513,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_fwd_stats", &conv_fwd_stats);
   m.def("conv_dgrad", &conv_dgrad);
+  m.def("conv_dgrad",
+        [](at::Tensor dy, at::Tensor w, std::vector<int64_t> xs, int64_t sh,
+           int64_t sw, int64_t ph, int64_t pw) {
+          return conv_dgrad(dy, w, xs, sh, sw, ph, pw, c10::nullopt);
+        });
   m.def("conv_wgrad", &conv_wgrad);
   m.def("conv_fwd_smallc", &conv_fwd_smallc);
   m.def("conv_wgrad_smallc", &conv_wgrad_smallc);

@@ -115,6 +115,10 @@ void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
                   int Q, int sh, int sw, int ph, int pw, int relu, int dtype,
                   hipStream_t s);
 // dgrad consumes w_t [C,R,S,K] (CRSK) built by ddl_transpose_krsc_crsk.
+void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
+                          const void* accum, int N, int H, int W, int C,
+                          int K, int R, int S, int P, int Q, int sh, int sw,
+                          int ph, int pw, int dtype, hipStream_t s);
 void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
                     int W, int C, int K, int R, int S, int P, int Q, int sh,
                     int sw, int ph, int pw, int dtype, hipStream_t s);

@@ -56,7 +56,8 @@ template <typename T, int MODE, bool DBUF, int TBM = 128, int TBN = 128>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
-    int Nc, int relu, DgradPhase gp, float* __restrict__ stats_ws) {
+    int Nc, int relu, DgradPhase gp, float* __restrict__ stats_ws,
+    const T* __restrict__ accum) {
   constexpr int AROWS = TBM / 32;     // staged A rows per thread
   constexpr int BROWS = TBN / 32;     // staged B rows per thread
   constexpr int ATB = TBM * 128;      // A tile bytes (128 B rows)
@@ -387,6 +388,7 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
           if (m >= M) continue;
           float f = acc[mr][nr][reg] + bv;
           if (relu) f = fmaxf(f, 0.f);
+          if (accum != nullptr) f += to_f32(accum[out_row_elem(m) + n]);
           out[out_row_elem(m) + n] = f;
         }
       }
@@ -427,8 +429,19 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     const int ub = (unit % UPR) * 8;
     const int64_t m = tile_m + row;
     if (m < M && ub < nvalid) {
-      *reinterpret_cast<uint4*>(out + out_row_elem(m) + tile_n + ub) =
-          *reinterpret_cast<const uint4*>(etile + row * TBN + ub);
+      uint4 u = *reinterpret_cast<const uint4*>(etile + row * TBN + ub);
+      const int64_t off = out_row_elem(m) + tile_n + ub;
+      if (accum != nullptr) {
+        // dgrad residual-join fusion: dx += the identity-path gradient
+        // here instead of a separate full-tensor autograd add
+        uint4 a = *reinterpret_cast<const uint4*>(accum + off);
+        T* ue = reinterpret_cast<T*>(&u);
+        const T* ae = reinterpret_cast<const T*>(&a);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ue[j] = from_f32<T>(to_f32(ue[j]) + to_f32(ae[j]));
+      }
+      *reinterpret_cast<uint4*>(out + off) = u;
     }
   }
   if (!emit_stats) return;
@@ -525,11 +538,11 @@ void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
 #define RUN(T, MODE, DB)                                                       \
   conv_igemm_k<T, MODE, DB><<<(unsigned)grid, 256, 0, s>>>(                    \
       (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp,    \
-      stats_ws)
+      stats_ws, nullptr)
 #define RUN64(T, MODE)                                                         \
   conv_igemm_k<T, MODE, false, 256, 64><<<(unsigned)grid, 256, 0, s>>>(        \
       (const T*)x, (const T*)w, (const T*)bias, (T*)y, d, M, Nc, relu, gp,    \
-      stats_ws)
+      stats_ws, nullptr)
 #define SEL(T, MODE)                                                           \
   if (n64) RUN64(T, MODE);                                                     \
   else if (dbuf) RUN(T, MODE, true);                                           \
@@ -555,6 +568,16 @@ int64_t ddl_conv_fwd_stats_chunks(int N, int P, int Q, int C, int K, int R,
 void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
                     int W, int C, int K, int R, int S, int P, int Q, int sh,
                     int sw, int ph, int pw, int dtype, hipStream_t s) {
+  ddl_conv_dgrad_accum(dy, w_t, dx, nullptr, N, H, W, C, K, R, S, P, Q, sh,
+                       sw, ph, pw, dtype, s);
+}
+
+// dx = conv_dgrad(dy, w) + accum  (accum nullable; the residual-join
+// gradient add fused into the epilogue)
+void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
+                          const void* accum, int N, int H, int W, int C,
+                          int K, int R, int S, int P, int Q, int sh, int sw,
+                          int ph, int pw, int dtype, hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int Nc = C;
   const bool dbuf = conv_use_dbuf(K, R * S);
@@ -566,10 +589,12 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
     DgradPhase gp{};
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD, DB><<<(unsigned)grid, 256, 0, s>>>(                   \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr, \
+      (const T*)accum)
 #define RUN64(T)                                                               \
   conv_igemm_k<T, DGRAD, false, 256, 64><<<(unsigned)grid, 256, 0, s>>>(       \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr, \
+      (const T*)accum)
     if (dtype == 0) { if (n64) RUN64(float); else if (dbuf) RUN(float, true); else RUN(float, false); }
     else if (dtype == 1) { if (n64) RUN64(bf16); else if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
     else if (dtype == 2) { if (n64) RUN64(f16); else if (dbuf) RUN(f16, true); else RUN(f16, false); }
@@ -594,7 +619,8 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
       const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
 #define RUN(T, DB)                                                             \
   conv_igemm_k<T, DGRAD_S2, DB><<<(unsigned)grid, 256, 0, s>>>(                \
-      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr)
+      (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp, nullptr, \
+      (const T*)accum)
       if (dtype == 0) { if (dbuf) RUN(float, true); else RUN(float, false); }
       else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
       else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }

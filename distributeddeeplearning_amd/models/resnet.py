@@ -14,6 +14,7 @@ from typing import List, Type, Union
 
 import torch.nn as nn
 
+from ..ops import functional as DF
 from ..ops.modules import (
     BatchNorm2d,
     Conv2d,
@@ -46,8 +47,12 @@ class BasicBlock(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = self.downsample(x) if self.downsample is not None else x
-        out = self.bn1(self.conv1(x), relu=True)
+        # fork-conv fusion: x gets ONE autograd consumer, so the residual
+        # join's gradient add folds into conv1's dgrad epilogue
+        ident, out = DF.fork_conv2d(x, self.conv1)
+        identity = (self.downsample(ident) if self.downsample is not None
+                    else ident)
+        out = self.bn1(out, relu=True)
         out = self.bn2(self.conv2(out), relu=True, residual=identity)
         return out
 
@@ -68,8 +73,10 @@ class Bottleneck(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = self.downsample(x) if self.downsample is not None else x
-        out = self.bn1(self.conv1(x), relu=True)
+        ident, out = DF.fork_conv2d(x, self.conv1)
+        identity = (self.downsample(ident) if self.downsample is not None
+                    else ident)
+        out = self.bn1(out, relu=True)
         out = self.bn2(self.conv2(out), relu=True)
         out = self.bn3(self.conv3(out), relu=True, residual=identity)
         return out

@@ -16,6 +16,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -137,6 +139,77 @@ def conv2d_stats(x, weight, bias=None, stride=1, padding=0):
                                       False, True)
         return y, (partials if partials.numel() > 0 else None)
     return F.conv2d(x, weight, bias, stride, padding), None
+
+
+class _ForkConvFn(torch.autograd.Function):
+    """(identity, y, partials) = (x, conv(x), bn-partials).
+
+    Routing the residual fork THROUGH the conv's autograd Function gives
+    the block input a single consumer, so backward receives both the
+    conv-path gradient (dy) and the identity-path gradient (dident) and
+    fuses ``dx = conv_dgrad(dy, w) + dident`` in the dgrad epilogue —
+    removing the separate full-activation autograd add at every residual
+    join (~5.5 ms/step at ResNet-50 b1024, the largest single
+    at::native elementwise cost)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding, emit_stats):
+        ext = load_kernels_ext()
+        x = to_channels_last(x)
+        weight = to_channels_last(weight)
+        if emit_stats:
+            y, partials = ext.conv_fwd_stats(x, weight, None, stride[0],
+                                             stride[1], padding[0],
+                                             padding[1], False)
+        else:
+            y = ext.conv_fwd(x, weight, None, stride[0], stride[1],
+                             padding[0], padding[1], False)
+            partials = torch.empty(0, dtype=torch.float32, device=x.device)
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.padding = padding
+        ident = x.view_as(x)
+        ctx.mark_non_differentiable(partials)
+        return ident, y, partials
+
+    @staticmethod
+    def backward(ctx, dident, dy, _dpartials):
+        ext = load_kernels_ext()
+        x, weight = ctx.saved_tensors
+        dy = to_channels_last(dy)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            acc = to_channels_last(dident) if dident is not None else None
+            dx = ext.conv_dgrad(dy, weight, list(x.shape), ctx.stride[0],
+                                ctx.stride[1], ctx.padding[0],
+                                ctx.padding[1], acc)
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv_wgrad(dy, x, list(weight.shape), ctx.stride[0],
+                                ctx.stride[1], ctx.padding[0],
+                                ctx.padding[1])
+        return dx, dw, None, None, None
+
+
+_FUSE_RESJOIN = os.environ.get("DDL_FUSE_RESJOIN", "1") == "1"
+
+
+def fork_conv2d(x, conv_module):
+    """Residual-block entry: returns (identity_source, conv_output) with
+    the join-gradient add fused into the conv's dgrad (GPU training path;
+    everything else falls back to the plain fork)."""
+    run_fused = (_FUSE_RESJOIN and x.is_cuda and conv_module.training
+                 and torch.is_grad_enabled()
+                 and x.shape[1] % (4 if x.dtype == torch.float32 else 8) == 0)
+    if not run_fused:
+        return x, conv_module(x)
+    emit = (conv_module.emit_bn_stats and conv_module.training
+            and x.dtype != torch.float32)
+    ident, y, partials = _ForkConvFn.apply(
+        x, conv_module.weight, conv_module.stride, conv_module.padding,
+        emit)
+    if partials.numel() > 0:
+        y._ddl_bn_partials = partials
+    return ident, y
 
 
 # --------------------------------------------------------------------------

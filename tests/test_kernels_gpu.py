@@ -353,3 +353,24 @@ def test_conv_bn_fusion_module_training_step():
     gu = torch.autograd.grad(yu.float().sum(), [x2, conv_u.weight])
     assert relerr(gf[0], gu[0]) < 2e-2
     assert relerr(gf[1], gu[1]) < 2e-2
+
+
+def test_conv_dgrad_accum_fusion():
+    """dgrad epilogue accumulate: conv_dgrad(dy, w, accum=a) equals
+    conv_dgrad(dy, w) + a (the fused residual-join gradient add)."""
+    ext = _ext()
+    dy = cl(mk((4, 128, 28, 28), seed=51).to(DEV))
+    w = cl(mk((128, 64, 3, 3), seed=52).to(DEV) * 0.1)
+    a = cl(mk((4, 64, 28, 28), seed=53).to(DEV))
+    xs = [4, 64, 28, 28]
+    base = ext.conv_dgrad(dy, w, xs, 1, 1, 1, 1)
+    fused = ext.conv_dgrad(dy, w, xs, 1, 1, 1, 1, a)
+    ref = (base.float() + a.float()).bfloat16()
+    assert relerr(fused, ref) < 2e-3
+    # strided phase-decomposed path too
+    dy2 = cl(mk((4, 128, 14, 14), seed=54).to(DEV))
+    a2 = cl(mk((4, 64, 28, 28), seed=55).to(DEV))
+    base2 = ext.conv_dgrad(dy2, w, xs, 2, 2, 1, 1)
+    fused2 = ext.conv_dgrad(dy2, w, xs, 2, 2, 1, 1, a2)
+    ref2 = (base2.float() + a2.float()).bfloat16()
+    assert relerr(fused2, ref2) < 2e-3
