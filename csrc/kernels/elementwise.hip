@@ -184,6 +184,59 @@ __global__ void im2col_k(const T* __restrict__ x, T* __restrict__ out,
   }
 }
 
+// Row-staged im2col: one block per (n, p) output row-group stages the R
+// input rows it needs into LDS (vectorized 16 B loads; W*C is a multiple
+// of 8 for the stem's 224*3 so rows stay 16 B aligned) and gathers the
+// (r,s,c) taps from LDS.  The per-element global-gather kernel above is
+// instruction/L1-transaction bound (8 scalar 2 B loads per 16 B unit,
+// VALUBusy ~56%) at ~5x its bandwidth roofline; staging turns the global
+// traffic into clean 16 B streams.
+template <typename T>
+__global__ __launch_bounds__(256) void im2col_rows_k(
+    const T* __restrict__ x, T* __restrict__ out,
+    const int* __restrict__ tbl, int N, int H, int W, int C, int R, int P,
+    int Q, int sh, int sw, int ph, int pw, int kred_total, int ldout) {
+  __shared__ __align__(16) char smem[48 * 1024];
+  T* slab = reinterpret_cast<T*>(smem);
+  const int n = blockIdx.x / P, p = blockIdx.x % P;
+  const int h0 = p * sh - ph;
+  const int rowlen = W * C;  // elements per input row
+  constexpr int EPU = 16 / (int)sizeof(T);
+  // stage R rows (out-of-image rows as zeros)
+  const int units_per_row = rowlen / EPU;
+  for (int u = threadIdx.x; u < R * units_per_row; u += 256) {
+    const int r = u / units_per_row;
+    const int uu = u % units_per_row;
+    const int hi = h0 + r;
+    uint4 v = uint4{0, 0, 0, 0};
+    if (hi >= 0 && hi < H)
+      v = *reinterpret_cast<const uint4*>(
+          x + ((int64_t)n * H + hi) * rowlen + uu * EPU);
+    *reinterpret_cast<uint4*>(slab + r * rowlen + uu * EPU) = v;
+  }
+  __syncthreads();
+  const int nunits = ldout / EPU;
+  const int64_t m_base = ((int64_t)n * P + p) * Q;
+  for (int i = threadIdx.x; i < Q * nunits; i += 256) {
+    const int q = i / nunits;
+    const int ug = i % nunits;
+    const int w0 = q * sw - pw;
+    T e[EPU] = {};
+#pragma unroll
+    for (int j = 0; j < EPU; ++j) {
+      const int kred = ug * EPU + j;
+      if (kred < kred_total) {
+        const int t = tbl[kred];
+        const int wi = w0 + ((t >> 10) & 1023);
+        if (wi >= 0 && wi < W)
+          e[j] = slab[(t >> 20) * rowlen + wi * C + (t & 1023)];
+      }
+    }
+    *reinterpret_cast<uint4*>(out + (m_base + q) * ldout + ug * EPU) =
+        *reinterpret_cast<const uint4*>(e);
+  }
+}
+
 // zero-pad the K x kred weight matrix into K x ldout
 template <typename T>
 __global__ void pad_cols_k(const T* __restrict__ w, T* __restrict__ out,
@@ -275,9 +328,25 @@ void ddl_cast_f32_to(const float* src, void* dst, int64_t n, int dtype,
 void ddl_im2col(const void* x, void* out, const int* tbl, int N, int H,
                 int W, int C, int R, int S, int P, int Q, int sh, int sw,
                 int ph, int pw, int ldout, int dtype, hipStream_t s) {
+  int kred_total = R * S * C;
+  const int esz = dtype == 0 ? 4 : 2;
+  const int epu = 16 / esz;
+  // row-staged variant: the block's R input rows must fit 48 KiB LDS and
+  // stay 16 B aligned (rowlen % EPU == 0)
+  const bool rows_ok = (int64_t)R * W * C * esz <= 48 * 1024 &&
+                       (W * C) % epu == 0 && (int64_t)N * P < (1u << 31);
+  if (rows_ok) {
+    int g = N * P;
+#define RUN(T)                                                                 \
+  im2col_rows_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, tbl, N, H, W, C,   \
+                                     R, P, Q, sh, sw, ph, pw, kred_total,     \
+                                     ldout)
+    switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+    return;
+  }
   int64_t total = (int64_t)N * P * Q * (ldout / 8);
   int g = (int)std::min<int64_t>(cdiv(total, 256), 8192);
-  int kred_total = R * S * C;
 #define RUN(T)                                                                 \
   im2col_k<T><<<g, 256, 0, s>>>((const T*)x, (T*)out, tbl, N, H, W, C, P,     \
                                 Q, sh, sw, ph, pw, kred_total, ldout)
