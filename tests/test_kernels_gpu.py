@@ -374,3 +374,38 @@ def test_conv_dgrad_accum_fusion():
     fused2 = ext.conv_dgrad(dy2, w, xs, 2, 2, 1, 1, a2)
     ref2 = (base2.float() + a2.float()).bfloat16()
     assert relerr(fused2, ref2) < 2e-3
+
+
+def test_fork_conv_residual_fusion_grads():
+    """fork_conv2d (residual-join add fused into dgrad) produces the same
+    gradients as the plain fork + autograd add."""
+    from distributeddeeplearning_amd.models.resnet import Bottleneck
+    from distributeddeeplearning_amd.ops import functional as DF
+    from distributeddeeplearning_amd.ops.modules import cast_model_
+    torch.manual_seed(11)
+    blk_f = Bottleneck(256, 64).to(DEV)
+    blk_u = Bottleneck(256, 64).to(DEV)
+    blk_u.load_state_dict(blk_f.state_dict())
+    for b in (blk_f, blk_u):
+        cast_model_(b, torch.bfloat16)
+        b.train()
+    x = cl(mk((4, 256, 14, 14), seed=61).to(DEV)).requires_grad_(True)
+    x2 = x.detach().clone().requires_grad_(True)
+    yf = blk_f(x)
+    old = DF._FUSE_RESJOIN
+    DF._FUSE_RESJOIN = False
+    try:
+        yu = blk_u(x2)
+    finally:
+        DF._FUSE_RESJOIN = old
+    assert torch.equal(yf.float(), yu.float()), "forward changed by fusion"
+    dy = mk(yf.shape, seed=62).to(DEV)
+    yf.backward(cl(dy))
+    yu.backward(cl(dy))
+    # dx: fused add in f32 inside the epilogue vs bf16 autograd add —
+    # equal up to one rounding step
+    assert relerr(x.grad, x2.grad) < 2e-3
+    for (n1, p1), (n2, p2) in zip(blk_f.named_parameters(),
+                                  blk_u.named_parameters()):
+        assert n1 == n2
+        assert relerr(p1.grad, p2.grad) < 2e-3, n1
