@@ -72,10 +72,27 @@ class BatchNorm2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # python-side shadow: `buf += 1` on a GPU scalar is one kernel
+        # launch per BN per step (53/step on ResNet-50); count here and
+        # flush into the buffer whenever a state_dict is taken
+        self._nbt_pending = 0
+
+    def _flush_nbt(self):
+        if self._nbt_pending:
+            self.num_batches_tracked += self._nbt_pending
+            self._nbt_pending = 0
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        self._flush_nbt()
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        self._nbt_pending = 0
+        super()._load_from_state_dict(*args, **kwargs)
 
     def forward(self, x, relu: bool = False, residual: Optional[torch.Tensor] = None):
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt_pending += 1
         partials = getattr(x, "_ddl_bn_partials", None)
         if partials is not None:
             del x._ddl_bn_partials
