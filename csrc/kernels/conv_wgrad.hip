@@ -14,6 +14,9 @@ namespace {
 
 enum WMode { TILED = 0, FLAT = 1 };
 
+typedef __attribute__((ext_vector_type(4))) short s16x4_t;
+typedef __attribute__((address_space(3))) s16x4_t* as3_s16x4p;
+
 // SIMPLE = 1x1/stride-1/no-pad conv: the x operand is just row m of the
 // input (no im2col decode; the per-tile q/p/n divides were a hot cost).
 // TBM/TBN/NTHR: tile geometry.  (128,128,256) is the full 4-wave 2x2 tile;
@@ -23,8 +26,17 @@ enum WMode { TILED = 0, FLAT = 1 };
 // The 64^2 TILED/SIMPLE variants sit at ~200 VGPR, just past the 192
 // 2-waves/SIMD boundary; minWavesPerEU=2 squeezes them back under it
 // (FLAT's im2col decode needs the registers more than the occupancy).
+// TR16 (2-byte dtypes, SBUF 128^2 only, DDL_WGRAD_TR16=1): stage the
+// tiles in their NATURAL [m][ch] layout — one plain ds_write_b128 per
+// 16 B load instead of the 8x4 register transpose (8 b64 writes + ~50
+// pack VALU) — and read MFMA fragments with gfx950's hardware transpose
+// read `ds_read_b64_tr_b16`.  Image: [ch-block cb][m-half][32 m][16 ch]
+// subtiles (32 B rows), slot base padded by cb*32 so the 8-lane b128
+// write groups spread banks, and rows XOR'd (m ^ ((m>>3)&1)<<2) so the
+// two 16-lane tr-read halves of a 32-lane service group (k-groups 8
+// rows apart) land on different banks.
 template <typename T, int MODE, bool SIMPLE, bool DBUF, int TBM = 128,
-          int TBN = 128, int NTHR = 256>
+          int TBN = 128, int NTHR = 256, bool TR16 = false>
 __global__ __launch_bounds__(NTHR, (NTHR == 64) ? 2 : 1)
 void conv_wgrad_k(
     const T* __restrict__ dy, const T* __restrict__ x,
@@ -32,7 +44,12 @@ void conv_wgrad_k(
     int64_t slab_len, FastDiv fdq, FastDiv fdpq, int ntiles, int chunks,
     int xcd_map) {
   static_assert(TBM == TBN, "square tiles only");
-  constexpr int ATILE = TBM * 128;  // bytes per operand tile (128 B rows)
+  static_assert(!TR16 || (sizeof(T) == 2 && !DBUF),
+                "TR16: 2-byte dtypes, single-buffer only");
+  // TR16 operand tile: TBM/16 ch-blocks x 2 m-halves slots x 1 KiB
+  // + cb padding
+  constexpr int ATILE =
+      TR16 ? (TBM / 16) * 2 * 1024 + (TBM / 16) * 32 : TBM * 128;
   // (tile, chunk) decode.  xcd_map=1: co-locate a chunk's tiles on one XCD
   // (block b runs on XCD b%8, MI355X_MICROARCH.md §Workgroup dispatch —
   // affinity used for speed only).  The tiles of one chunk sweep the same
@@ -147,8 +164,22 @@ void conv_wgrad_k(
       }
   };
 
+  // TR16 subtile addressing: slot (cb, mh) holds logical m-rows
+  // mh*32..+31 of channels cb*16..+15; 32 B rows, row-XOR for tr reads.
+  auto tr16_off = [&](int cb, int mh, int m31, int chbyte) {
+    const int pm = m31 ^ (((m31 >> 3) & 1) << 2);
+    return (cb * 2 + mh) * 1024 + cb * 32 + pm * 32 + chbyte;
+  };
   auto transpose_write = [&](char* lds, const uint4 (&v)[4], int sm4p) {
-    if constexpr (sizeof(T) == 2) {
+    if constexpr (TR16) {
+      const int cb = su >> 1, half = su & 1;
+#pragma unroll
+      for (int mm = 0; mm < 4; ++mm) {
+        const int m = sm4p + mm;
+        *reinterpret_cast<uint4*>(
+            lds + tr16_off(cb, m >> 5, m & 31, half * 16)) = v[mm];
+      }
+    } else if constexpr (sizeof(T) == 2) {
       const uint16_t* e0 = reinterpret_cast<const uint16_t*>(&v[0]);
       const uint16_t* e1 = reinterpret_cast<const uint16_t*>(&v[1]);
       const uint16_t* e2 = reinterpret_cast<const uint16_t*>(&v[2]);
@@ -183,7 +214,48 @@ void conv_wgrad_k(
   auto mfma_tile = [&](const char* la, const char* lb) {
 #pragma unroll
     for (int kk = 0; kk < ConvElems<T>::KSTEPS; ++kk) {
-      if constexpr (sizeof(T) == 2) {
+      if constexpr (TR16) {
+        // hardware transpose read: each 16-lane group reads a
+        // [4 m][16 ch] block; lane l gets ch (l&15), m = 8*(l>>4)+j.
+        // Two reads (+4 m rows) build the 8-deep mfma fragment.
+        const int lx = lane & 15;
+        const int lg = lane >> 4;
+        const int row0 = 8 * lg + (lx >> 2);
+        const int pr0 = row0 ^ (((row0 >> 3) & 1) << 2);
+        const int row1 = row0 + 4;
+        const int pr1 = row1 ^ (((row1 >> 3) & 1) << 2);
+        const int lane_off = (lx & 3) * 8;
+        typename MfmaFrag<T>::type a[4], b[4];
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr) {
+          const int cb = (wr + mr * 16) >> 4;
+          const char* sb = la + (cb * 2 + kk) * 1024 + cb * 32 + lane_off;
+          s16x4_t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (as3_s16x4p)(sb + pr0 * 32));
+          s16x4_t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (as3_s16x4p)(sb + pr1 * 32));
+          s16x4_t* h = reinterpret_cast<s16x4_t*>(&a[mr]);
+          h[0] = r0;
+          h[1] = r1;
+        }
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr) {
+          const int cb = (wc + nr * 16) >> 4;
+          const char* sb = lb + (cb * 2 + kk) * 1024 + cb * 32 + lane_off;
+          s16x4_t r0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (as3_s16x4p)(sb + pr0 * 32));
+          s16x4_t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (as3_s16x4p)(sb + pr1 * 32));
+          s16x4_t* h = reinterpret_cast<s16x4_t*>(&b[nr]);
+          h[0] = r0;
+          h[1] = r1;
+        }
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+          for (int nr = 0; nr < 4; ++nr)
+            acc[mr][nr] = mfma_16x16x32<T>(a[mr], b[nr], acc[mr][nr]);
+      } else if constexpr (sizeof(T) == 2) {
         const int fcolb = kk * 64 + (lane >> 4) * 16;
         typename MfmaFrag<T>::type a[4], b[4];
 #pragma unroll
@@ -420,7 +492,13 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   const bool simple = (R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
                        pw == 0 && !flat);
   const char* env = getenv("DDL_WGRAD_DBUF");
-  const bool dbuf = env ? env[0] == '1' : true;  // measured: dbuf wins here
+  bool dbuf = env ? env[0] == '1' : true;  // measured: dbuf wins here
+  const char* te = getenv("DDL_WGRAD_TR16");
+  // measured 2-2.5x over the register-transpose staging on every
+  // 128^2-tile shape (l3.c1 218->649 TF); FLAT keeps the old path (its
+  // im2col decode is register-bound)
+  const bool tr16 = (te ? te[0] == '1' : true) && dtype != 0 && !flat;
+  if (tr16) dbuf = false;
   FastDiv fdq = make_fastdiv((uint32_t)Q);
   FastDiv fdpq = make_fastdiv((uint32_t)P * Q);
 #define RUN(T, MODE, SIMP, DB)                                                 \
@@ -431,14 +509,31 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
   conv_wgrad_k<T, MODE, SIMP, false, 64, 64, 64><<<grid, 64, 0, s>>>(          \
       (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
       fdpq, ntiles, (int)pl.chunks, xcd_map)
+#define RUN64TR(T, MODE, SIMP)                                                 \
+  conv_wgrad_k<T, MODE, SIMP, false, 64, 64, 64, true><<<grid, 64, 0, s>>>(    \
+      (const T*)dy, (const T*)x, dw_ws, d, M, pl.chunk_rows, slab_len, fdq,    \
+      fdpq, ntiles, (int)pl.chunks, xcd_map)
+#define RUNTR(T, MODE, SIMP)                                                   \
+  conv_wgrad_k<T, MODE, SIMP, false, 128, 128, 256, true>                      \
+      <<<grid, 256, 0, s>>>((const T*)dy, (const T*)x, dw_ws, d, M,            \
+                            pl.chunk_rows, slab_len, fdq, fdpq, ntiles,        \
+                            (int)pl.chunks, xcd_map)
+#define SELTR(T, MODE, SIMP)                                                   \
+  if (tr16) RUNTR(T, MODE, SIMP);                                              \
+  else if (dbuf) RUN(T, MODE, SIMP, true);                                     \
+  else RUN(T, MODE, SIMP, false)
 #define SEL(T, MODE, SIMP)                                                     \
+  if (pl.tile == 64) { if (tr16) RUN64TR(T, MODE, SIMP);                       \
+                       else RUN64(T, MODE, SIMP); }                            \
+  else SELTR(T, MODE, SIMP)
+#define SELF32(T, MODE, SIMP)                                                  \
   if (pl.tile == 64) RUN64(T, MODE, SIMP);                                     \
   else if (dbuf) RUN(T, MODE, SIMP, true);                                     \
   else RUN(T, MODE, SIMP, false)
   if (dtype == 0) {
-    if (flat) { SEL(float, FLAT, false); }
-    else if (simple) { SEL(float, TILED, true); }
-    else { SEL(float, TILED, false); }
+    if (flat) { SELF32(float, FLAT, false); }
+    else if (simple) { SELF32(float, TILED, true); }
+    else { SELF32(float, TILED, false); }
   } else if (dtype == 1) {
     if (flat) { SEL(bf16, FLAT, false); }
     else if (simple) { SEL(bf16, TILED, true); }
@@ -448,7 +543,11 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
     else if (simple) { SEL(f16, TILED, true); }
     else { SEL(f16, TILED, false); }
   }
+#undef SELF32
 #undef SEL
+#undef SELTR
+#undef RUNTR
+#undef RUN64TR
 #undef RUN64
 #undef RUN
   int g = (int)cdiv(slab_len, 64);
