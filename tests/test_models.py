@@ -83,3 +83,22 @@ def test_vgg16_backward_cpu():
     loss.backward()
     assert torch.isfinite(loss)
     assert all(p.grad is not None for p in m.parameters())
+
+
+def test_bn_num_batches_tracked_flushes_to_state_dict():
+    """The python-side shadow counter (no per-step GPU kernel) must land
+    in state_dict / checkpoints exactly like the eager buffer did."""
+    import torch
+    from distributeddeeplearning_amd.ops.modules import BatchNorm2d
+    bn = BatchNorm2d(8)
+    bn.train()
+    x = torch.randn(4, 8, 5, 5)
+    for _ in range(3):
+        bn(x)
+    sd = bn.state_dict()
+    assert int(sd["num_batches_tracked"]) == 3
+    bn2 = BatchNorm2d(8)
+    bn2.load_state_dict(sd)
+    bn2.train()
+    bn2(x)
+    assert int(bn2.state_dict()["num_batches_tracked"]) == 4
