@@ -102,3 +102,20 @@ def test_bn_num_batches_tracked_flushes_to_state_dict():
     bn2.train()
     bn2(x)
     assert int(bn2.state_dict()["num_batches_tracked"]) == 4
+
+
+def test_fork_conv_and_stats_cpu_fallbacks():
+    """CPU paths: conv2d_stats -> (y, None); fork_conv2d -> plain fork."""
+    import torch
+    from distributeddeeplearning_amd.ops import functional as DF
+    from distributeddeeplearning_amd.ops.modules import Conv2d
+    conv = Conv2d(8, 16, 3, padding=1, emit_bn_stats=True)
+    conv.train()
+    x = torch.randn(2, 8, 6, 6, requires_grad=True)
+    y, partials = DF.conv2d_stats(x, conv.weight, None, 1, 1)
+    assert partials is None
+    ident, out = DF.fork_conv2d(x, conv)
+    assert ident is x
+    assert torch.allclose(out, y, atol=1e-6)
+    out.sum().backward()
+    assert x.grad is not None and conv.weight.grad is not None
