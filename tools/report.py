@@ -42,11 +42,15 @@ def profile_table(db_path, out=None):
 def scaling_table(paths):
     rows = []
     for p in paths:
-        with open(p) as f:
-            for line in f:
-                line = line.strip()
-                if line.startswith("{") and '"images/sec' in line:
-                    rows.append(json.loads(line))
+        text = open(p).read().strip()
+        if text.startswith("["):  # whole-file JSON array (driver SCALE file)
+            rows.extend(r for r in json.loads(text)
+                        if isinstance(r, dict) and "n_gpus" in r)
+            continue
+        for line in text.splitlines():
+            line = line.strip()
+            if line.startswith("{") and '"images/sec' in line:
+                rows.append(json.loads(line))
     rows.sort(key=lambda r: r["n_gpus"])
     if not rows:
         print("no bench JSON lines found")
