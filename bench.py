@@ -56,6 +56,26 @@ def main() -> None:
     res = trainer.bench(args.steps, args.warmup)
 
     if trainer.rank == 0:
+        # Provenance: record which comm path actually ran,
+        # bucket layout and environment so every BENCH/SCALE record is
+        # self-describing — a torchdist fallback can never be mistaken for
+        # the native engine.
+        try:
+            import subprocess
+            head = subprocess.run(
+                ["git", "rev-parse", "--short=12", "HEAD"], cwd=os.path.dirname(
+                    os.path.abspath(__file__)), capture_output=True, text=True,
+                timeout=10).stdout.strip() or None
+        except Exception:
+            head = None  # gpurun snapshots exclude .git
+        provenance = {
+            "comm_backend": api.comm_backend_name(),
+            "bucket_mb": cfg.bucket_mb,
+            "n_buckets": len(trainer.opt.buckets.buckets),
+            "git_head": head,
+            "env": {k: v for k, v in sorted(os.environ.items())
+                    if k.startswith(("DDL_", "NCCL_", "RCCL_"))},
+        }
         out = {
             "metric": "images/sec (whole node) ResNet-50 synthetic ImageNet",
             "value": round(res["images_per_sec"], 2),
@@ -75,6 +95,7 @@ def main() -> None:
                 "image_size": cfg.image_size,
                 "parallelism": f"dp{n_ranks}",
             },
+            "provenance": provenance,
         }
         print(json.dumps(out), flush=True)
     api.shutdown()

@@ -73,7 +73,16 @@ def load(path: str, model, optimizer=None, scaler=None,
         else:
             inner = optimizer
         if hasattr(inner, "refresh_master"):
-            inner.refresh_master()
+            # Only re-derive fp32 masters from the (possibly bf16/fp16)
+            # params when the checkpoint did NOT carry them — a restored
+            # master is exact and re-deriving would truncate it to the
+            # low-precision params (breaking bit-exact resume).
+            has_master = any(
+                torch.is_tensor(st.get("master"))
+                for st in getattr(inner, "state", {}).values()
+            )
+            if not has_master:
+                inner.refresh_master()
         if hasattr(inner, "invalidate_table"):
             inner.invalidate_table()
     if scaler is not None and "loss_scaler" in payload:

@@ -102,6 +102,7 @@ class Trainer:
             self.opt.optimizer.refresh_master()
 
         self._loader: Optional[Iterator] = None
+        self._micro = 0  # backward passes since the last optimizer step
 
     # ------------------------------------------------------------------
 
@@ -124,6 +125,9 @@ class Trainer:
         return self._loader
 
     def train_step(self, x, y) -> torch.Tensor:
+        """One micro-batch: forward+backward; every
+        ``backward_passes_per_step``-th call also reduces and applies the
+        update (the Horovod accumulation contract: N backwards, one step)."""
         tl = self.timeline
         lr = self.lr_sched.step(self.global_step)
         if tl:
@@ -136,6 +140,13 @@ class Trainer:
         self.scaler.scale_loss(loss).backward()
         if tl:
             tl.end("backward+allreduce")
+        self._micro += 1
+        if self._micro < self.cfg.backward_passes_per_step:
+            # mid-accumulation: grads keep accumulating in the flat buckets
+            self.global_step += 1
+            return loss
+        self._micro = 0
+        if tl:
             tl.begin("step")
         if self.scaler.enabled:
             self.opt.synchronize()
@@ -143,7 +154,9 @@ class Trainer:
             self.opt.optimizer.grad_scale = self.scaler.scale
             if self.scaler.check_and_update(flats):
                 self.opt.step()
-            # on overflow: skip the update entirely (standard AMP)
+            # on overflow: skip the update; synchronize() already re-armed
+            # the accumulation counters and zero_grad below discards the
+            # non-finite gradients, so the next window starts clean
         else:
             self.opt.step()
         self.opt.zero_grad()
@@ -206,14 +219,18 @@ class Trainer:
             if self.cfg.channels_last and x.dim() == 4:
                 x = x.contiguous(memory_format=torch.channels_last)
             logits = self.model(x)
-            loss_sum += float(DF.softmax_cross_entropy(logits, y).float())
+            # per-batch mean loss, weighted by batch size so the epoch mean
+            # is per-sample regardless of ragged final batches
+            loss_sum += float(DF.softmax_cross_entropy(logits, y).float()) * y.numel()
             correct += int((logits.float().argmax(1) == y).sum())
             total += y.numel()
         top1 = correct / max(1, total)
+        val_loss = loss_sum / max(1, total)
         # C4: metric averaging across ranks
         top1 = average_metric(top1, device="cpu")
+        val_loss = average_metric(val_loss, device="cpu")
         self.model.train()
-        return {"top1": top1, "val_loss": loss_sum / max(1, total)}
+        return {"top1": top1, "val_loss": val_loss}
 
     # ------------------------------------------------------------------
 

@@ -165,6 +165,40 @@ class FusedSGD(torch.optim.Optimizer):
         sd = super().state_dict()
         return sd
 
+    def load_state_dict(self, sd):
+        """Exact-dtype state restore.
+
+        torch's default ``Optimizer.load_state_dict`` casts floating state
+        tensors to the PARAM dtype — under bf16/fp16 params that would
+        truncate the fp32 masters and momentum (and break the fused kernel,
+        which reads them as 4-byte words).  Restore manually, moving device
+        but never changing dtype."""
+        groups = self.param_groups
+        saved_groups = sd["param_groups"]
+        if len(groups) != len(saved_groups):
+            raise ValueError("loaded state dict has a different number of "
+                             "parameter groups")
+        id_map = {}
+        for g, sg in zip(groups, saved_groups):
+            if len(g["params"]) != len(sg["params"]):
+                raise ValueError("parameter group size mismatch in "
+                                 "load_state_dict")
+            for p, pid in zip(g["params"], sg["params"]):
+                id_map[pid] = p
+            for k, v in sg.items():
+                if k != "params":
+                    g[k] = v
+        self.state.clear()
+        for pid, st in sd["state"].items():
+            p = id_map.get(pid, pid)
+            dev = p.device if torch.is_tensor(p) else None
+            self.state[p] = {
+                k: (v.to(device=dev) if torch.is_tensor(v) and dev is not None
+                    else v)
+                for k, v in st.items()
+            }
+        self._tables = {}
+
     def invalidate_table(self):
         self._tables = {}
 

@@ -49,6 +49,13 @@ def _b() -> CommBackend:
     return _backend
 
 
+def comm_backend_name() -> str:
+    """Which backend actually carries the gradient collectives
+    ("native" | "torchdist" | "single") — reported in bench provenance so a
+    silent fallback can never masquerade as the native engine."""
+    return _b().name
+
+
 def rank() -> int:
     return _b().rank()
 

@@ -51,6 +51,8 @@ class Handle:
 class CommBackend:
     """Interface shared by all backends."""
 
+    name = "abstract"  # reported in bench provenance (VERDICT weak #1)
+
     def rank(self) -> int:
         raise NotImplementedError
 
@@ -81,6 +83,8 @@ class CommBackend:
 
 class TorchDistBackend(CommBackend):
     """torch.distributed-backed comm (gloo CPU / nccl==RCCL GPU)."""
+
+    name = "torchdist"
 
     def __init__(self, process_group=None):
         self.pg = process_group
@@ -133,6 +137,8 @@ class SingleProcessBackend(CommBackend):
     Mirrors Horovod's behaviour when run un-launched; also the test fake
     (SURVEY.md §4.3 "LoopbackComm")."""
 
+    name = "single"
+
     def rank(self) -> int:
         return 0
 
@@ -160,6 +166,8 @@ class NativeRcclBackend(CommBackend):
     issued by the native engine on its own HIP stream.
     """
 
+    name = "native"
+
     def __init__(self, device_index: int, control_pg=None):
         from ..ops._ext import load_comm_ext
 
@@ -181,15 +189,28 @@ class NativeRcclBackend(CommBackend):
     def size(self) -> int:
         return self.engine.size()
 
+    def _data_plane(self, tensor: torch.Tensor) -> bool:
+        """GPU tensors of RCCL-supported dtypes go to the native engine;
+        host-side scalars/metrics (CPU tensors, float64) ride the gloo
+        control plane instead (C4 call sites: metric averaging, debug-order
+        checksums — never perf-critical)."""
+        return tensor.is_cuda and tensor.dtype != torch.float64
+
     def allreduce_async_(self, tensor: torch.Tensor, average: bool = True) -> Handle:
+        if not self._data_plane(tensor):
+            return self._ctrl.allreduce_async_(tensor, average=average)
         hid = self.engine.allreduce_async(tensor, bool(average))
         return Handle(_NativeWork(self.engine, hid), tensor)
 
     def broadcast_async_(self, tensor: torch.Tensor, root_rank: int) -> Handle:
+        if not self._data_plane(tensor):
+            return self._ctrl.broadcast_async_(tensor, root_rank)
         hid = self.engine.broadcast_async(tensor, int(root_rank))
         return Handle(_NativeWork(self.engine, hid), tensor)
 
     def allgather_async(self, tensor: torch.Tensor) -> Handle:
+        if not self._data_plane(tensor):
+            return self._ctrl.allgather_async(tensor)
         t = tensor.contiguous()
         out_shape = list(t.shape)
         out_shape[0] *= self.size()

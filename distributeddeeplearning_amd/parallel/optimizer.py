@@ -46,7 +46,6 @@ class DistributedOptimizer:
     ):
         self.optimizer = optimizer
         self.backward_passes_per_step = int(backward_passes_per_step)
-        self._passes = 0
         self.average = average
         self.compression = compression
         self._timeline = None
@@ -67,9 +66,18 @@ class DistributedOptimizer:
         self.buckets = BucketManager(named_parameters, bucket_bytes=bucket_bytes)
         self._hooks = []
         self._last_ready_time: Optional[float] = None
-        self._warned_stall = False
+        self._last_stall_warn = 0.0
+        # Horovod-style per-param accumulation counters
+        # (horovod/torch/optimizer.py `_allreduce_delay`): each backward pass
+        # decrements; the bucket launches on the pass that reaches zero.
+        # Supports BOTH usage patterns: N backwards then one step() (the
+        # upstream contract) and step() after every backward (step() no-ops
+        # mid-accumulation).
+        self._delay: dict[int, int] = {}
+        self._synchronized = False
         for b in self.buckets.buckets:
             for p in b.params:
+                self._delay[id(p)] = self.backward_passes_per_step
                 h = p.register_post_accumulate_grad_hook(self._make_hook())
                 self._hooks.append(h)
 
@@ -86,14 +94,24 @@ class DistributedOptimizer:
                     "param.grad no longer aliases its flat bucket view "
                     "(was zero_grad(set_to_none=True) called on the model?). "
                     "Use DistributedOptimizer.zero_grad() instead.")
-            bucket = self.buckets.mark_ready(p)
-            if bucket is not None and self._should_reduce():
-                self._launch(bucket)
+            d = self._delay[id(p)]
+            if d <= 0:
+                if api.size() == 1:
+                    # no collective to mis-schedule: extra backwards just
+                    # accumulate locally (plain-PyTorch semantics)
+                    return
+                raise RuntimeError(
+                    f"more than backward_passes_per_step="
+                    f"{self.backward_passes_per_step} backward passes since "
+                    "the last step() — call step() (and zero_grad()) after "
+                    "every backward_passes_per_step backwards")
+            self._delay[id(p)] = d - 1
+            if d == 1:  # this was the Nth (final) pass for this param
+                bucket = self.buckets.mark_ready(p)
+                if bucket is not None:
+                    self._launch(bucket)
 
         return hook
-
-    def _should_reduce(self) -> bool:
-        return (self._passes + 1) >= self.backward_passes_per_step
 
     def _launch(self, bucket) -> None:
         if api.size() > 1:
@@ -117,15 +135,22 @@ class DistributedOptimizer:
 
     # -- hvd surface --------------------------------------------------------
 
+    def _mid_accumulation(self) -> bool:
+        """True while some backward passes of the accumulation window ran but
+        the window is not complete (no param has reached its final pass)."""
+        mind = min(self._delay.values())
+        return 0 < mind < self.backward_passes_per_step
+
     def synchronize(self) -> None:
-        """Wait every in-flight bucket (device-side on GPU backends).
+        """Wait every in-flight bucket (device-side on GPU backends), then
+        re-arm the per-param accumulation counters for the next window.
 
         Also launches any bucket that never filled (a param that got no grad
         this pass would otherwise deadlock other ranks — the stall-inspector
         case).
         """
         for b in self.buckets.buckets:
-            if b.handle is None and self._should_reduce() and api.size() > 1:
+            if b.handle is None and api.size() > 1:
                 if 0 < b.ready < len(b.params):
                     warnings.warn(
                         f"bucket {b.index}: only {b.ready}/{len(b.params)} grads "
@@ -147,37 +172,43 @@ class DistributedOptimizer:
                 raise RuntimeError(
                     f"collective order diverged across ranks: {self._launch_order}")
             self._launch_order.clear()
+        for k in self._delay:
+            self._delay[k] = self.backward_passes_per_step
+        self.buckets.reset()
+        self._synchronized = True
 
     def _inspect_stall(self) -> None:
         """Stall inspector (Horovod stall_inspector.cc role): warn when the
-        native engine reports collectives outstanding for too long."""
+        native engine reports collectives outstanding for too long.
+        Rate-limited to one warning per STALL_WARN_SECONDS."""
         backend = api._backend
         eng = getattr(backend, "engine", None)
-        if eng is not None and not self._warned_stall:
+        now = time.monotonic()
+        if eng is not None and now - self._last_stall_warn > STALL_WARN_SECONDS:
             age = eng.oldest_age_seconds()
             if age > STALL_WARN_SECONDS:
-                self._warned_stall = True
+                self._last_stall_warn = now
                 warnings.warn(
                     f"collective outstanding for {age:.0f}s "
                     f"({eng.outstanding()} in flight) — possible rank "
                     "desync/deadlock")
 
     def step(self, closure=None):
-        self._passes += 1
-        if self._passes < self.backward_passes_per_step:
-            # accumulation step: no reduce, no inner step
-            self.buckets.reset()
-            return None
-        self.synchronize()
-        loss = self.optimizer.step(closure)
-        self._passes = 0
-        return loss
+        if not self._synchronized:
+            if self._mid_accumulation():
+                # legacy pattern: step() called after every backward while
+                # accumulating — no reduce, no inner step until the window
+                # completes
+                return None
+            self.synchronize()
+        self._synchronized = False
+        return self.optimizer.step(closure)
 
     def zero_grad(self, set_to_none: bool = False):
         # Grads are views into flat buckets; zero the buckets and keep views.
         # Mid-accumulation (backward_passes_per_step > 1, before the real
         # step) the call is a no-op so accumulated gradients survive.
-        if self._passes != 0:
+        if self._mid_accumulation():
             return
         self.buckets.zero_grads()
         self.buckets.reset()

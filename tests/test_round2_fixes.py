@@ -1,0 +1,217 @@
+"""Round-2 regression tests for the ADVICE r1 findings.
+
+Covers: the Horovod accumulation contract (N backwards then one step),
+over-backward detection, fp16-overflow recovery mid-accumulation, exact
+fp32-master restore under low-precision params, and CPU-metric reduction
+routing.
+"""
+import copy
+import os
+
+import pytest
+import torch
+
+from tests.dist_helpers import run_distributed
+
+
+def _mlp(seed=7):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.Tanh(), torch.nn.Linear(32, 4))
+
+
+# -- ADVICE high: Horovod pattern (N backwards, ONE step) -------------------
+
+def _w_horovod_pattern(rank, world):
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("torchdist")
+    model = _mlp(seed=100)
+    api.broadcast_parameters(model.state_dict(), root_rank=0)
+    opt = DistributedOptimizer(
+        FusedSGD(model.parameters(), lr=0.05, momentum=0.9),
+        named_parameters=model.named_parameters(), bucket_bytes=2048,
+        backward_passes_per_step=2)
+    for step in range(2):
+        for micro in range(2):  # N backwards ...
+            g = torch.Generator().manual_seed(step * 10 + micro)
+            x = torch.randn(8, 16, generator=g)
+            y = torch.randn(8, 4, generator=g)
+            xr = x[rank * 4:(rank + 1) * 4]
+            yr = y[rank * 4:(rank + 1) * 4]
+            torch.nn.functional.mse_loss(model(xr), yr).backward()
+        opt.step()        # ... then ONE step (upstream Horovod contract)
+        opt.zero_grad()
+    params = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    return {"params": params}
+
+
+def _horovod_pattern_reference():
+    """Single process doing the same accumulated full-batch steps."""
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    model = _mlp(seed=100)
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    for step in range(2):
+        for micro in range(2):
+            g = torch.Generator().manual_seed(step * 10 + micro)
+            x = torch.randn(8, 16, generator=g)
+            y = torch.randn(8, 4, generator=g)
+            torch.nn.functional.mse_loss(model(x), y).backward()
+        # DP averages grads over ranks; each rank's quarter-batch mse_loss
+        # means over 4 rows, the reference over 8 → scale grads by 1/1
+        # (mean over 4 rows on 2 ranks averaged == mean over 8 rows).
+        opt.step()
+        opt.zero_grad()
+    return torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+
+def test_horovod_accumulation_pattern_world2():
+    """N backwards then one step must reduce on the Nth
+    backward and apply exactly once — ranks identical and equal to the
+    single-process accumulated trajectory."""
+    res = run_distributed(_w_horovod_pattern, world=2)
+    assert torch.equal(res[0]["params"], res[1]["params"]), "ranks diverged"
+    ref = _horovod_pattern_reference()
+    assert torch.allclose(res[0]["params"], ref, atol=1e-5), \
+        float((res[0]["params"] - ref).abs().max())
+
+
+def _w_over_backward(rank, world):
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("torchdist")
+    model = _mlp(seed=3)
+    opt = DistributedOptimizer(
+        FusedSGD(model.parameters(), lr=0.05),
+        named_parameters=model.named_parameters(),
+        backward_passes_per_step=1)
+    x = torch.randn(4, 16)
+    model(x).sum().backward()
+    try:
+        model(x).sum().backward()  # 2nd backward without step(): must raise
+    except RuntimeError as e:
+        return {"raised": "backward_passes_per_step" in str(e)}
+    return {"raised": False}
+
+
+def test_over_backward_raises_multirank():
+    """Extra backward passes beyond the declared window must fail loudly
+    multi-rank instead of silently skipping the reduction."""
+    res = run_distributed(_w_over_backward, world=2)
+    assert all(r["raised"] for r in res)
+
+
+# -- ADVICE medium: fp16 overflow mid-accumulation must not wedge -----------
+
+def test_fp16_overflow_recovers():
+    from distributeddeeplearning_amd.engine.loss_scaler import DynamicLossScaler
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import DistributedOptimizer, api
+    api.init("single")
+    torch.manual_seed(0)
+    model = _mlp(seed=1)
+    opt = DistributedOptimizer(
+        FusedSGD(model.parameters(), lr=0.05, momentum=0.9),
+        named_parameters=model.named_parameters(),
+        backward_passes_per_step=2)
+    scaler = DynamicLossScaler(init_scale=1024.0, enabled=True)
+    before = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+
+    def window(inject_inf):
+        for _ in range(2):
+            x = torch.randn(4, 16)
+            scaler.scale_loss(model(x).sum()).backward()
+        opt.synchronize()
+        if inject_inf:
+            opt.buckets.buckets[0].flat.view(-1)[0] = float("inf")
+        flats = [b.flat for b in opt.buckets.buckets]
+        opt.optimizer.grad_scale = scaler.scale
+        stepped = False
+        if scaler.check_and_update(flats):
+            opt.step()
+            stepped = True
+        opt.zero_grad()
+        return stepped
+
+    # overflow window: no update, scale backs off, buckets cleared
+    assert window(inject_inf=True) is False
+    after_skip = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    assert torch.equal(before, after_skip), "params changed on skipped step"
+    assert scaler.scale == 512.0
+    for b in opt.buckets.buckets:
+        assert b.flat.abs().sum() == 0, "inf grads survived the skip"
+
+    # next window must proceed normally (the r1 bug wedged here forever)
+    assert window(inject_inf=False) is True
+    after_good = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    assert not torch.equal(before, after_good), "recovery step did not apply"
+
+
+# -- ADVICE low: exact fp32-master restore under bf16 params ----------------
+
+def test_resume_restores_fp32_master_exactly(tmp_path):
+    from distributeddeeplearning_amd.engine import checkpoint as ckpt
+    from distributeddeeplearning_amd.ops.optim import FusedSGD
+    from distributeddeeplearning_amd.parallel import api
+    api.init("single")
+    torch.manual_seed(5)
+    model = torch.nn.Linear(8, 8).to(torch.bfloat16)
+    opt = FusedSGD(model.parameters(), lr=0.01, momentum=0.9)
+    for _ in range(3):
+        model(torch.randn(4, 8, dtype=torch.bfloat16)).sum().backward()
+        opt.step()
+        model.zero_grad()
+    masters = {id(p): opt.state[p]["master"].clone()
+               for g in opt.param_groups for p in g["params"]}
+    path = str(tmp_path / "ck.pt")
+    ckpt.save(path, model, opt, epoch=0, step=3)
+
+    model2 = torch.nn.Linear(8, 8).to(torch.bfloat16)
+    opt2 = FusedSGD(model2.parameters(), lr=0.01, momentum=0.9)
+    ckpt.load(path, model2, opt2)
+    for (p1, m1), p2 in zip(
+            [(p, masters[id(p)]) for g in opt.param_groups for p in g["params"]],
+            [p for g in opt2.param_groups for p in g["params"]]):
+        st = opt2.state[p2]
+        assert st["master"].dtype == torch.float32, \
+            "master was cast to the param dtype on load"
+        assert torch.equal(st["master"], m1), "master not bit-exact after resume"
+        assert st["momentum_buffer"].dtype == torch.float32
+
+    # continuing both for one identical step stays bit-identical
+    x = torch.randn(4, 8, dtype=torch.bfloat16)
+    model(x).sum().backward()
+    opt.step()
+    model2(x).sum().backward()
+    opt2.step()
+    for pa, pb in zip(model.parameters(), model2.parameters()):
+        assert torch.equal(pa, pb)
+
+
+# -- ADVICE medium: CPU metric reduction under the native backend ----------
+# (the native engine itself is GPU-only; here we verify the routing
+# predicate that sends host scalars to the control plane)
+
+def test_native_backend_routes_host_tensors_to_control_plane():
+    from distributeddeeplearning_amd.parallel.backend import NativeRcclBackend
+    pred = NativeRcclBackend._data_plane
+    class _B:  # no engine needed: predicate is pure
+        pass
+    cpu64 = torch.zeros(1, dtype=torch.float64)
+    cpu32 = torch.zeros(1, dtype=torch.float32)
+    assert pred(_B(), cpu64) is False
+    assert pred(_B(), cpu32) is False  # CPU tensors never hit RCCL
+    # (CUDA tensors can't be constructed here; the GPU test suite covers
+    # the positive case via the full training path)
+
+
+def _w_eval_metrics(rank, world):
+    from distributeddeeplearning_amd.parallel import api
+    from distributeddeeplearning_amd.engine.metrics import average_metric
+    api.init("torchdist")
+    return {"avg": average_metric(float(rank))}
+
+
+def test_average_metric_world2():
+    res = run_distributed(_w_eval_metrics, world=2)
+    assert all(abs(r["avg"] - 0.5) < 1e-12 for r in res)
