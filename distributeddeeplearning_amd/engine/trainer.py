@@ -132,14 +132,18 @@ class Trainer:
         lr = self.lr_sched.step(self.global_step)
         if tl:
             tl.begin("forward")
+            tl.begin_device("forward")
         logits = self.model(x)
         loss = DF.softmax_cross_entropy(logits, y, self.cfg.label_smoothing)
         if tl:
             tl.end("forward")
+            tl.end_device("forward")
             tl.begin("backward+allreduce")
+            tl.begin_device("backward+allreduce")
         self.scaler.scale_loss(loss).backward()
         if tl:
             tl.end("backward+allreduce")
+            tl.end_device("backward+allreduce")
         self._micro += 1
         if self._micro < self.cfg.backward_passes_per_step:
             # mid-accumulation: grads keep accumulating in the flat buckets
@@ -148,6 +152,7 @@ class Trainer:
         self._micro = 0
         if tl:
             tl.begin("step")
+            tl.begin_device("step")
         if self.scaler.enabled:
             self.opt.synchronize()
             flats = [b.flat for b in self.opt.buckets.buckets]
@@ -162,6 +167,7 @@ class Trainer:
         self.opt.zero_grad()
         if tl:
             tl.end("step")
+            tl.end_device("step")
             if self.global_step % 50 == 0:
                 tl.save()
         self.global_step += 1

@@ -172,10 +172,24 @@ class NativeRcclBackend(CommBackend):
         from ..ops._ext import load_comm_ext
 
         self._ctrl = TorchDistBackend(control_pg)
-        self._ext = load_comm_ext()  # raises if the .so is missing
+        # Pre-init consensus: every rank must have the extension loaded and
+        # (on rank 0) a unique id BEFORE anyone enters ncclCommInitRank —
+        # a rank that fails after its peers enter the blocking init would
+        # hang the job instead of falling back.
+        err: Optional[Exception] = None
         uid = None
-        if self._ctrl.rank() == 0:
-            uid = self._ext.get_unique_id()
+        try:
+            self._ext = load_comm_ext()  # raises if the .so is missing
+            if self._ctrl.rank() == 0:
+                uid = self._ext.get_unique_id()
+        except Exception as e:  # noqa: BLE001
+            err = e
+        ok = torch.tensor([0 if err is not None else 1], dtype=torch.int64)
+        torch_dist.all_reduce(ok, op=torch_dist.ReduceOp.MIN, group=control_pg)
+        if int(ok.item()) != 1:
+            raise RuntimeError(
+                f"native RCCL engine unavailable on at least one rank "
+                f"(local error: {err!r})")
         obj = [uid]
         torch_dist.broadcast_object_list(obj, src=0, group=control_pg)
         uid = obj[0]
@@ -276,15 +290,30 @@ def create_backend(comm: Optional[str] = None) -> CommBackend:
     elif comm == "native-or-torchdist":
         dev = _default_device_index()
         torch.cuda.set_device(dev)
+        native = None
+        err = None
         try:
-            return NativeRcclBackend(dev)
+            native = NativeRcclBackend(dev)
         except Exception as e:  # noqa: BLE001 — any init failure
-            import warnings
-            warnings.warn(
-                f"native RCCL engine init failed ({e!r}); falling back to "
-                "torch.distributed nccl (still RCCL, without the native "
-                "side-stream engine)")
-            return TorchDistBackend()
+            err = e
+        # Consensus fallback: the backend choice must be IDENTICAL on every
+        # rank — a per-rank fallback (some ranks native, some torchdist)
+        # would deadlock the first collective.  NativeRcclBackend's own
+        # pre-init consensus covers the common failure points; this second
+        # vote covers a rank failing inside engine construction itself.
+        flag = torch.tensor([1 if native is not None else 0], dtype=torch.int64)
+        torch_dist.all_reduce(flag, op=torch_dist.ReduceOp.MIN)
+        if int(flag.item()) == 1:
+            return native
+        import warnings
+        warnings.warn(
+            f"native RCCL engine init failed on at least one rank "
+            f"(local error: {err!r}); ALL ranks falling back to "
+            "torch.distributed nccl (still RCCL, without the native "
+            "side-stream engine). Set DDL_COMM=native to make this fatal.")
+        if native is not None:
+            native.shutdown()
+        return TorchDistBackend()
     elif comm == "torchdist":
         return TorchDistBackend()
     elif comm == "single":
