@@ -508,6 +508,19 @@ bool conv_use_dbuf(int cred, int rs_total) {
 
 }  // namespace
 
+// 8-phase 256^2 deep-pipeline variant (conv_igemm8.hip)
+extern "C" int ddl_conv8_applicable(int dtype, int64_t M, int Nc);
+extern "C" void ddl_conv8_fwd(const void* x, const void* w, const void* bias,
+                              void* y, float* stats_ws, int N, int H, int W,
+                              int C, int K, int R, int S, int P, int Q,
+                              int sh, int sw, int ph, int pw, int relu,
+                              int dtype, hipStream_t s);
+extern "C" void ddl_conv8_dgrad(const void* dy, const void* w_t, void* dx,
+                                const void* accum, int N, int H, int W, int C,
+                                int K, int R, int S, int P, int Q, int sh,
+                                int sw, int ph, int pw, int dtype,
+                                hipStream_t s);
+
 extern "C" {
 
 void ddl_conv_fwd(const void* x, const void* w, const void* bias, void* y,
@@ -528,6 +541,11 @@ void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
   const int64_t M = (int64_t)N * P * Q;
   const int Nc = K;
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
+  if (!flat && ddl_conv8_applicable(dtype, M, Nc)) {
+    ddl_conv8_fwd(x, w, bias, y, stats_ws, N, H, W, C, K, R, S, P, Q, sh,
+                  sw, ph, pw, relu, dtype, s);
+    return;
+  }
   const bool n64 = !flat && conv_use_n64(Nc, M);
   const int64_t grid =
       n64 ? cdiv(M, 256) * cdiv(Nc, 64) : cdiv(M, BM) * cdiv(Nc, BN);
@@ -561,6 +579,7 @@ int64_t ddl_conv_fwd_stats_chunks(int N, int P, int Q, int C, int K, int R,
                                   int S, int dtype) {
   const int64_t M = (int64_t)N * P * Q;
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
+  if (!flat && ddl_conv8_applicable(dtype, M, K)) return cdiv(M, 256);
   const bool n64 = !flat && conv_use_n64(K, M);
   return n64 ? cdiv(M, 256) : cdiv(M, BM);
 }
@@ -580,6 +599,15 @@ void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
                           int ph, int pw, int dtype, hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int Nc = C;
+  {
+    // per-phase M for strided dgrad (one launch per (h%sh, w%sw) phase)
+    const int64_t Mck = (int64_t)N * cdiv(H, sh) * cdiv(W, sw);
+    if (ddl_conv8_applicable(dtype, Mck, Nc)) {
+      ddl_conv8_dgrad(dy, w_t, dx, accum, N, H, W, C, K, R, S, P, Q, sh, sw,
+                      ph, pw, dtype, s);
+      return;
+    }
+  }
   const bool dbuf = conv_use_dbuf(K, R * S);
   if (sh == 1 && sw == 1) {
     const int64_t M = (int64_t)N * H * W;
