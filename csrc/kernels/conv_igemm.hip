@@ -509,7 +509,7 @@ bool conv_use_dbuf(int cred, int rs_total) {
 }  // namespace
 
 // 8-phase 256^2 deep-pipeline variant (conv_igemm8.hip)
-extern "C" int ddl_conv8_applicable(int dtype, int64_t M, int Nc);
+extern "C" int ddl_conv8_applicable(int dtype, int64_t M, int Nc, int rs_total, int Cred);
 extern "C" void ddl_conv8_fwd(const void* x, const void* w, const void* bias,
                               void* y, float* stats_ws, int N, int H, int W,
                               int C, int K, int R, int S, int P, int Q,
@@ -541,7 +541,7 @@ void ddl_conv_fwd_stats(const void* x, const void* w, const void* bias,
   const int64_t M = (int64_t)N * P * Q;
   const int Nc = K;
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
-  if (!flat && ddl_conv8_applicable(dtype, M, Nc)) {
+  if (!flat && ddl_conv8_applicable(dtype, M, Nc, R * S, C)) {
     ddl_conv8_fwd(x, w, bias, y, stats_ws, N, H, W, C, K, R, S, P, Q, sh,
                   sw, ph, pw, relu, dtype, s);
     return;
@@ -579,7 +579,7 @@ int64_t ddl_conv_fwd_stats_chunks(int N, int P, int Q, int C, int K, int R,
                                   int S, int dtype) {
   const int64_t M = (int64_t)N * P * Q;
   const bool flat = (C % (dtype == 0 ? 4 : 8)) != 0;
-  if (!flat && ddl_conv8_applicable(dtype, M, K)) return cdiv(M, 256);
+  if (!flat && ddl_conv8_applicable(dtype, M, K, R * S, C)) return cdiv(M, 256);
   const bool n64 = !flat && conv_use_n64(K, M);
   return n64 ? cdiv(M, 256) : cdiv(M, BM);
 }
@@ -602,7 +602,8 @@ void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
   {
     // per-phase M for strided dgrad (one launch per (h%sh, w%sw) phase)
     const int64_t Mck = (int64_t)N * cdiv(H, sh) * cdiv(W, sw);
-    if (ddl_conv8_applicable(dtype, Mck, Nc)) {
+    const int rsp = (sh == 1 && sw == 1) ? R * S : cdiv(R, sh) * cdiv(S, sw);
+    if (ddl_conv8_applicable(dtype, Mck, Nc, rsp, K)) {
       ddl_conv8_dgrad(dy, w_t, dx, accum, N, H, W, C, K, R, S, P, Q, sh, sw,
                       ph, pw, dtype, s);
       return;

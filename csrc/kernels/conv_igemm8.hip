@@ -412,16 +412,24 @@ __global__ __launch_bounds__(512, 2) void conv_igemm8_k(
 
 extern "C" {
 
-// Applicability: 2-byte dtype, vectorizable C (handled by caller), and a
-// grid that fills the chip at 1 workgroup/CU.  DDL_CONV_8PH=0 disables.
-int ddl_conv8_applicable(int dtype, int64_t M, int Nc) {
+// Applicability: 2-byte dtype, vectorizable C (handled by caller), a deep
+// enough reduce pipeline and a grid that fills the chip at 1 WG/CU.
+// Measured A/B (profiles/conv8_ab.md): at batch 256 the schedule wins
+// from ~12 tiles of K-depth (l3.c2: +31% fwd / +35% dgrad); at batch
+// 1024 the 128^2 kernel's larger grid already covers latency and only
+// the deepest loops still win (l4.c2 72 tiles: +15% fwd / +7% dgrad).
+// Route the depth the win is measured at on the headline batch; losses
+// appear when the 256^2 grid underfills 256 CUs at 1 WG/CU or Nc < 192.
+// DDL_CONV_8PH=0 disables, =2 forces (for sweeps).
+int ddl_conv8_applicable(int dtype, int64_t M, int Nc, int rs_total,
+                         int Cred) {
   if (dtype == 0) return 0;
   const char* env = getenv("DDL_CONV_8PH");
   if (env && env[0] == '0') return 0;
+  if (env && env[0] == '2') return 1;
+  const int64_t tiles = (int64_t)rs_total * cdiv(Cred, 64);
   const int64_t grid = cdiv(M, BM8) * cdiv(Nc, BN8);
-  if (grid < 64) return 0;
-  if (Nc <= 64 && !(env && env[0] == '2')) return 0;  // n64 tile wins there
-  return 1;
+  return tiles >= 48 && grid >= 160 && Nc >= 192;
 }
 
 void ddl_conv8_fwd(const void* x, const void* w, const void* bias, void* y,
