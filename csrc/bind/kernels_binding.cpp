@@ -233,6 +233,43 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
   return dx;
 }
 
+// dgrad that also emits the PRECEDING BatchNorm's backward stats partials
+// from its epilogue (the bwd analog of conv_fwd_stats): returns
+// (dx, partials[chunks*2*C]).  2-byte dtypes, channel-aligned only —
+// caller checks via the same gate as conv_dgrad.
+std::tuple<at::Tensor, at::Tensor> conv_dgrad_bnstats(
+    at::Tensor dy, at::Tensor w, std::vector<int64_t> x_shape, int64_t sh,
+    int64_t sw, int64_t ph, int64_t pw, at::Tensor bnx, at::Tensor bn_mean,
+    at::Tensor bn_rstd, at::Tensor bn_w, at::Tensor bn_b) {
+  check_cl(dy, "dy"); check_cl(w, "w"); check_cl(bnx, "bnx");
+  int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
+  int K = w.size(0), R = w.size(2), S = w.size(3);
+  int P = dy.size(2), Q = dy.size(3);
+  const int dt = dt_code(dy);
+  TORCH_CHECK(dt != 0 && C % 8 == 0,
+              "dgrad+bnstats fusion needs 2-byte dtype and C % 8 == 0");
+  TORCH_CHECK(bnx.sizes() == at::IntArrayRef(x_shape) &&
+                  bnx.scalar_type() == dy.scalar_type(),
+              "bnx must match the conv input shape/dtype");
+  TORCH_CHECK(bn_mean.scalar_type() == at::kFloat &&
+              bn_rstd.scalar_type() == at::kFloat);
+  TORCH_CHECK(bn_w.scalar_type() == dy.scalar_type() &&
+              bn_b.scalar_type() == dy.scalar_type());
+  auto w_t = at::empty({(int64_t)C * R * S * K}, w.options());
+  ddl_transpose_krsc_crsk(w.data_ptr(), w_t.data_ptr(), K, R * S, C,
+                          dt_code(w), cur_stream());
+  auto dx = empty_nhwc(N, C, H, W, dy.options());
+  const int64_t chunks = ddl_conv_dgrad_bnstats_chunks(N, H, W, C, K, sh, sw);
+  auto partials = at::empty({chunks * 2 * C}, dy.options().dtype(at::kFloat));
+  ddl_conv_dgrad_bnstats(dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(),
+                         bnx.data_ptr(), bn_mean.data_ptr<float>(),
+                         bn_rstd.data_ptr<float>(), bn_w.data_ptr(),
+                         bn_b.data_ptr(), partials.data_ptr<float>(), N, H,
+                         W, C, K, R, S, P, Q, sh, sw, ph, pw, dt,
+                         cur_stream());
+  return {dx, partials};
+}
+
 at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
                       std::vector<int64_t> w_shape, int64_t sh, int64_t sw,
                       int64_t ph, int64_t pw) {
@@ -333,7 +370,7 @@ at::Tensor bn_relu_infer(at::Tensor x, at::Tensor weight, at::Tensor bias,
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
     at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
     at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
-    bool has_residual) {
+    bool has_residual, c10::optional<at::Tensor> pre_partials) {
   check_cl(dy, "dy"); check_cl(x, "x");
   int64_t C = x.size(1);
   int64_t M = x.numel() / C;
@@ -342,13 +379,30 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
   // relu mask source: with a residual, y != bn(x) so the mask must come
   // from y; without one, recompute (w*xhat+b > 0) and skip reading y.
   const int relu_mode = relu ? (has_residual ? 1 : 2) : 0;
-  auto sums = at::empty({2 * C}, x.options().dtype(at::kFloat));
-  int chunks = ddl_bn_stats_chunks(M, C);
-  auto partials = at::empty({(int64_t)chunks * 2 * C}, x.options().dtype(at::kFloat));
-  ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu_mode,
-                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                   weight.data_ptr(), bias.data_ptr(), M, C, dt,
-                   sums.data_ptr<float>(), partials.data_ptr<float>(), chunks, s);
+  at::Tensor sums;
+  if (pre_partials.has_value() && pre_partials->numel() > 0) {
+    // stats partials already emitted by the consuming conv's dgrad
+    // epilogue (conv_dgrad_bnstats): reduce them, skip the stats pass
+    TORCH_CHECK(relu_mode == 2, "pre-computed bwd partials imply the "
+                "recomputed-mask (no-residual) path");
+    const int64_t pchunks = pre_partials->numel() / (2 * C);
+    TORCH_CHECK(pchunks * 2 * C == pre_partials->numel(),
+                "bwd partials shape mismatch");
+    auto fold_ws = at::empty({(2 + 512) * C}, x.options().dtype(at::kFloat));
+    ddl_bn_reduce_sums(pre_partials->data_ptr<float>(),
+                       fold_ws.data_ptr<float>(), (int)pchunks, C, s);
+    sums = fold_ws.narrow(0, 0, 2 * C);
+  } else {
+    sums = at::empty({2 * C}, x.options().dtype(at::kFloat));
+    int chunks = ddl_bn_stats_chunks(M, C);
+    auto partials = at::empty({(int64_t)chunks * 2 * C},
+                              x.options().dtype(at::kFloat));
+    ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu_mode,
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     weight.data_ptr(), bias.data_ptr(), M, C, dt,
+                     sums.data_ptr<float>(), partials.data_ptr<float>(),
+                     chunks, s);
+  }
   auto dx = at::empty_like(x);
   at::Tensor dres;
   void* dres_p = nullptr;
@@ -513,6 +567,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd);
   m.def("conv_fwd_stats", &conv_fwd_stats);
   m.def("conv_dgrad", &conv_dgrad);
+  m.def("conv_dgrad_bnstats", &conv_dgrad_bnstats);
   m.def("conv_dgrad",
         [](at::Tensor dy, at::Tensor w, std::vector<int64_t> xs, int64_t sh,
            int64_t sw, int64_t ph, int64_t pw) {
@@ -532,6 +587,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         });
   m.def("bn_relu_infer", &bn_relu_infer);
   m.def("bn_relu_bwd", &bn_relu_bwd);
+  m.def("bn_relu_bwd",
+        [](at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
+           at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
+           bool has_residual) {
+          return bn_relu_bwd(dy, x, weight, bias, mean, rstd, y, relu,
+                             has_residual, c10::nullopt);
+        });
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("gap_fwd", &gap_fwd);

@@ -122,6 +122,22 @@ void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
 void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
                     int W, int C, int K, int R, int S, int P, int Q, int sh,
                     int sw, int ph, int pw, int dtype, hipStream_t s);
+// dgrad that also emits the PRECEDING BatchNorm's backward partials
+// [chunks][2][C] {sum(g*xhat), sum(g)} from its epilogue (2-byte dtypes;
+// relu_mode-2 mask recomputed from the BN's input bnx).  That BN then
+// skips its bn_bwd_stats pass.
+int64_t ddl_conv_dgrad_bnstats_chunks(int N, int H, int W, int C, int K,
+                                      int sh, int sw);
+void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
+                            const void* bnx, const float* bn_mean,
+                            const float* bn_rstd, const void* bn_w,
+                            const void* bn_b, float* bstats_ws, int N, int H,
+                            int W, int C, int K, int R, int S, int P, int Q,
+                            int sh, int sw, int ph, int pw, int dtype,
+                            hipStream_t s);
+// reduce bwd partials [chunks][2][C] -> sums[2][C] (deterministic)
+void ddl_bn_reduce_sums(const float* partials, float* fold_ws, int chunks,
+                        int C, hipStream_t s);
 // wgrad: fp32 per-chunk slabs in dw_ws (ddl_conv_wgrad_ws_elems floats,
 // no pre-zero needed) reduced+cast into dw (KRSC, dtype).  Deterministic.
 int64_t ddl_conv_wgrad_ws_elems(int N, int C, int K, int R, int S, int P,

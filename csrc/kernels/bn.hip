@@ -425,6 +425,24 @@ void ddl_bn_reduce_finalize(const float* partials, float* fold_ws, int chunks,
       running_var, mean_out, rstd_out, update_running);
 }
 
+// reduce-only (no finalize): bwd partials [chunks][2][C] -> sums in
+// fold_ws[0..2C).  fold_ws must hold (2 + 512) * C floats for the
+// two-stage path (stage slab [256][2][C]).
+void ddl_bn_reduce_sums(const float* partials, float* fold_ws, int chunks,
+                        int C, hipStream_t s) {
+  float* sums = fold_ws;
+  if (chunks > 512) {
+    const int fold = 256;
+    float* stage = fold_ws + 2 * (int64_t)C;
+    dim3 g(cdiv(C, 64), fold);
+    bn_fold_partials_k<<<g, 256, 0, s>>>(partials, stage, C, chunks, fold);
+    bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(stage, sums, C, fold);
+  } else {
+    bn_reduce_partials_k<<<cdiv(C, 64), 256, 0, s>>>(partials, sums, C,
+                                                     chunks);
+  }
+}
+
 void ddl_bn_apply(const void* x, void* y, const void* residual,
                   const float* mean, const float* rstd, const void* weight,
                   const void* bias, int64_t M, int C, int relu, int dtype,

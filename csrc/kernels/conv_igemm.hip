@@ -52,12 +52,25 @@ __device__ __forceinline__ void glds16(const void* gsrc, void* lds_base) {
 // puts the 4 waves in a 4x1 m-column over a 64-wide n-tile for Nc <= 64
 // outputs (ResNet layer1 / the im2col'd stem: a 128-wide n-tile there runs
 // 50% dead MFMA work).  SBUF only (A tile is 32 KiB at TBM=256).
-template <typename T, int MODE, bool DBUF, int TBM = 128, int TBN = 128>
+// bn_bwd fusion (DGRAD modes, 2-byte dtypes): when bstats_ws != nullptr the
+// epilogue also emits the PRECEDING BatchNorm's backward partials
+// [chunk][2][Nc] {sum(g*xhat), sum(g)} with g = this dgrad's output masked
+// by the BN's recomputed ReLU (w*xhat+b > 0, relu_mode 2) and
+// xhat = (bnx - mean)*rstd from the BN's own input — that BN then skips
+// its bn_bwd_stats full (dy, x) read pass (the backward analog of the
+// conv_fwd_stats fusion; docs/ROADMAP.md #8).
+template <typename T, int MODE, bool DBUF, int TBM = 128, int TBN = 128,
+          bool BSTATS = false>
 __global__ __launch_bounds__(256) void conv_igemm_k(
     const T* __restrict__ Asrc, const T* __restrict__ Bsrc,
     const T* __restrict__ bias, T* __restrict__ out, ConvDims d, int64_t M,
     int Nc, int relu, DgradPhase gp, float* __restrict__ stats_ws,
-    const T* __restrict__ accum) {
+    const T* __restrict__ accum, const T* __restrict__ bnx = nullptr,
+    const float* __restrict__ bn_mean = nullptr,
+    const float* __restrict__ bn_rstd = nullptr,
+    const T* __restrict__ bn_w = nullptr,
+    const T* __restrict__ bn_b = nullptr,
+    float* __restrict__ bstats_ws = nullptr, int bchunk_base = 0) {
   constexpr int AROWS = TBM / 32;     // staged A rows per thread
   constexpr int BROWS = TBN / 32;     // staged B rows per thread
   constexpr int ATB = TBM * 128;      // A tile bytes (128 B rows)
@@ -422,6 +435,23 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
   __syncthreads();
   const int nvalid = Nc - tile_n;
   constexpr int UPR = TBN / 8;  // 16 B units per etile row
+  // bn_bwd fusion per-thread state: this thread's column group is FIXED
+  // across store iterations (256 % UPR == 0), so the BN per-channel
+  // constants hoist out of the loop and the partial sums live in 16 VGPRs.
+  float bs0[8] = {}, bs1[8] = {};
+  float bmean[8], brstd[8], bw8[8], bb8[8];
+  if constexpr (BSTATS) {
+    const int cbase = tile_n + (threadIdx.x % UPR) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = cbase + j;
+      const bool ok = c < Nc;
+      bmean[j] = ok ? bn_mean[c] : 0.f;
+      brstd[j] = ok ? bn_rstd[c] : 0.f;
+      bw8[j] = ok ? to_f32(bn_w[c]) : 0.f;
+      bb8[j] = ok ? to_f32(bn_b[c]) : 0.f;
+    }
+  }
 #pragma unroll
   for (int it = 0; it < TBM * TBN / 8 / 256; ++it) {
     const int unit = threadIdx.x + it * 256;
@@ -441,7 +471,48 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
         for (int j = 0; j < 8; ++j)
           ue[j] = from_f32<T>(to_f32(ue[j]) + to_f32(ae[j]));
       }
+      if constexpr (BSTATS) {
+        // g = this dgrad row (the BN's dy), masked by the recomputed ReLU
+        uint4 bx = *reinterpret_cast<const uint4*>(bnx + off);
+        const T* be = reinterpret_cast<const T*>(&bx);
+        const T* ue = reinterpret_cast<const T*>(&u);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xh = (to_f32(be[j]) - bmean[j]) * brstd[j];
+          if (xh * bw8[j] + bb8[j] > 0.f) {
+            const float g = to_f32(ue[j]);
+            bs0[j] += g * xh;
+            bs1[j] += g;
+          }
+        }
+      }
       *reinterpret_cast<uint4*>(out + off) = u;
+    }
+  }
+  if constexpr (BSTATS) {
+    __syncthreads();
+    // fold the 256/UPR threads sharing each column group through LDS
+    float* bred = reinterpret_cast<float*>(smem);  // [256][8][2] = 16 KiB
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bred[(threadIdx.x * 8 + j) * 2 + 0] = bs0[j];
+      bred[(threadIdx.x * 8 + j) * 2 + 1] = bs1[j];
+    }
+    __syncthreads();
+    const int col = threadIdx.x;
+    if (col < TBN && tile_n + col < Nc) {
+      const int ccg = col >> 3, cj = col & 7;
+      float s0 = 0.f, s1 = 0.f;
+#pragma unroll 4
+      for (int r = 0; r < 256 / UPR; ++r) {
+        s0 += bred[((r * UPR + ccg) * 8 + cj) * 2 + 0];
+        s1 += bred[((r * UPR + ccg) * 8 + cj) * 2 + 1];
+      }
+      float* dst = bstats_ws +
+                   (int64_t)(bchunk_base + (int)(bid / nt_n)) * 2 * Nc +
+                   tile_n + col;
+      dst[0] = s0;
+      dst[Nc] = s1;
     }
   }
   if (!emit_stats) return;
@@ -633,6 +704,7 @@ void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
   }
   // strided dgrad: one dense launch per (h%sh, w%sw) phase
   for (int ph2 = 0; ph2 < sh; ++ph2) {
+    // (plain path — bn_bwd-stats fusion goes through ddl_conv_dgrad_bnstats)
     for (int pw2 = 0; pw2 < sw; ++pw2) {
       DgradPhase gp;
       gp.ph2 = ph2; gp.pw2 = pw2;
@@ -654,6 +726,74 @@ void ddl_conv_dgrad_accum(const void* dy, const void* w_t, void* dx,
       else if (dtype == 1) { if (dbuf) RUN(bf16, true); else RUN(bf16, false); }
       else if (dtype == 2) { if (dbuf) RUN(f16, true); else RUN(f16, false); }
 #undef RUN
+    }
+  }
+}
+
+
+// dgrad + BN-backward-stats fusion: dx = conv_dgrad(dy, w) and
+// bstats_ws[chunk][2][C] = the preceding BatchNorm's {sum(g*xhat), sum(g)}
+// partials (relu_mode-2 mask recomputed from its own input bnx).
+// 2-byte dtypes only; no accum (the fused residual-join path and this
+// fusion never coincide: residual-join BNs take the mode-1 mask).
+int64_t ddl_conv_dgrad_bnstats_chunks(int N, int H, int W, int C, int K,
+                                      int sh, int sw) {
+  if (sh == 1 && sw == 1) {
+    const int64_t M = (int64_t)N * H * W;
+    return conv_use_n64(C, M) ? cdiv(M, 256) : cdiv(M, BM);
+  }
+  int64_t chunks = 0;
+  for (int ph2 = 0; ph2 < sh; ++ph2)
+    for (int pw2 = 0; pw2 < sw; ++pw2) {
+      const int Hp = ph2 < H ? (H - ph2 + sh - 1) / sh : 0;
+      const int Wp = pw2 < W ? (W - pw2 + sw - 1) / sw : 0;
+      if (Hp && Wp) chunks += cdiv((int64_t)N * Hp * Wp, BM);
+    }
+  return chunks;
+}
+
+void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
+                            const void* bnx, const float* bn_mean,
+                            const float* bn_rstd, const void* bn_w,
+                            const void* bn_b, float* bstats_ws, int N, int H,
+                            int W, int C, int K, int R, int S, int P, int Q,
+                            int sh, int sw, int ph, int pw, int dtype,
+                            hipStream_t s) {
+  ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
+  const int Nc = C;
+  if (sh == 1 && sw == 1) {
+    const int64_t M = (int64_t)N * H * W;
+    const bool n64 = conv_use_n64(Nc, M);
+    const int64_t grid =
+        n64 ? cdiv(M, 256) * cdiv(Nc, 64) : cdiv(M, BM) * cdiv(Nc, BN);
+    DgradPhase gp{};
+#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD, false, 128, 128, true>                                      <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0)
+#define RUNB64(T)                                                                conv_igemm_k<T, DGRAD, false, 256, 64, true>                                       <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0)
+    if (dtype == 1) { if (n64) RUNB64(bf16); else RUNB(bf16); }
+    else { if (n64) RUNB64(f16); else RUNB(f16); }
+#undef RUNB64
+#undef RUNB
+    return;
+  }
+  int chunk_base = 0;
+  for (int ph2 = 0; ph2 < sh; ++ph2) {
+    for (int pw2 = 0; pw2 < sw; ++pw2) {
+      DgradPhase gp;
+      gp.ph2 = ph2; gp.pw2 = pw2;
+      gp.r0 = ((ph2 + ph) % sh + sh) % sh;
+      gp.s0 = ((pw2 + pw) % sw + sw) % sw;
+      gp.Sp = gp.s0 < S ? (S - gp.s0 + sw - 1) / sw : 0;
+      gp.Hp = ph2 < H ? (H - ph2 + sh - 1) / sh : 0;
+      gp.Wp = pw2 < W ? (W - pw2 + sw - 1) / sw : 0;
+      if (gp.Hp == 0 || gp.Wp == 0) continue;
+      gp.fdWp = make_fastdiv((uint32_t)gp.Wp);
+      gp.fdHpWp = make_fastdiv((uint32_t)gp.Hp * gp.Wp);
+      const int64_t M = (int64_t)N * gp.Hp * gp.Wp;
+      const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
+#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD_S2, false, 128, 128, true>                                   <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, chunk_base)
+      if (dtype == 1) RUNB(bf16); else RUNB(f16);
+#undef RUNB
+      chunk_base += (int)cdiv(M, BM);
     }
   }
 }

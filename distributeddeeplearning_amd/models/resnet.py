@@ -40,7 +40,9 @@ class BasicBlock(nn.Module):
     def __init__(self, inplanes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = conv3x3(inplanes, planes, stride)
-        self.bn1 = BatchNorm2d(planes)
+        # bn1's output has exactly one consumer (conv2): its backward
+        # stats come from conv2's dgrad epilogue (BnBwdFusion)
+        self.bn1 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv2 = conv3x3(planes, planes)
         self.bn2 = BatchNorm2d(planes)
         self.downsample = downsample
@@ -64,9 +66,11 @@ class Bottleneck(nn.Module):
     def __init__(self, inplanes, planes, stride=1, downsample=None):
         super().__init__()
         self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = BatchNorm2d(planes)
+        # bn1/bn2 each feed exactly one conv: their backward stats come
+        # from that conv's dgrad epilogue (BnBwdFusion)
+        self.bn1 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = BatchNorm2d(planes)
+        self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv3 = conv1x1(planes, planes * self.expansion)
         self.bn3 = BatchNorm2d(planes * self.expansion)
         self.downsample = downsample

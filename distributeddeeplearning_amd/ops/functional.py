@@ -38,9 +38,26 @@ def to_channels_last(t: torch.Tensor) -> torch.Tensor:
 # Conv2d (K1/K2/K3: implicit-GEMM NHWC, MFMA)
 # --------------------------------------------------------------------------
 
+class BnBwdFusion:
+    """Side-channel between a BatchNorm and the conv that consumes its
+    output: the conv's dgrad epilogue emits the BN's backward stats
+    partials (the bwd analog of conv_fwd_stats), so the BN skips its
+    bn_bwd_stats full (dy, x) pass.  The BN fills mean/rstd at forward;
+    the conv's backward fills partials; the BN's backward (which autograd
+    runs after the conv's) consumes them."""
+
+    __slots__ = ("mean", "rstd", "partials")
+
+    def __init__(self):
+        self.mean = None
+        self.rstd = None
+        self.partials = None
+
+
 class _Conv2dFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, stride, padding, relu, emit_stats=False):
+    def forward(ctx, x, weight, bias, stride, padding, relu, emit_stats=False,
+                bn_fuse=None):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         weight = to_channels_last(weight)
@@ -77,6 +94,8 @@ class _Conv2dFn(torch.autograd.Function):
         ctx.has_bias = bias is not None
         ctx.relu = relu
         ctx.emit_stats = emit_stats
+        # (holder, bnx, bn_w, bn_b) — read-only refs, no grads through here
+        ctx.bn_fuse = bn_fuse if not ctx.small_c else None
         if emit_stats:
             if partials is None:
                 partials = torch.empty(0, dtype=torch.float32,
@@ -105,9 +124,19 @@ class _Conv2dFn(torch.autograd.Function):
             if ctx.small_c:
                 raise RuntimeError("dgrad for C%8!=0 (stem) is not needed "
                                    "(input has no grad) and not implemented")
-            dx = ext.conv_dgrad(dy, weight, list(x.shape),
-                                ctx.stride[0], ctx.stride[1],
-                                ctx.padding[0], ctx.padding[1])
+            fuse = ctx.bn_fuse
+            if (fuse is not None and fuse[0].mean is not None
+                    and x.dtype != torch.float32):
+                holder, bnx, bn_w, bn_b = fuse
+                dx, bparts = ext.conv_dgrad_bnstats(
+                    dy, weight, list(x.shape), ctx.stride[0], ctx.stride[1],
+                    ctx.padding[0], ctx.padding[1], to_channels_last(bnx),
+                    holder.mean, holder.rstd, bn_w, bn_b)
+                holder.partials = bparts
+            else:
+                dx = ext.conv_dgrad(dy, weight, list(x.shape),
+                                    ctx.stride[0], ctx.stride[1],
+                                    ctx.padding[0], ctx.padding[1])
         if ctx.needs_input_grad[1]:
             if ctx.small_c:
                 dw = ext.conv_wgrad_smallc(dy, col, list(weight.shape),
@@ -118,25 +147,27 @@ class _Conv2dFn(torch.autograd.Function):
                                     ctx.padding[0], ctx.padding[1])
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.sum(dim=(0, 2, 3))
-        return dx, dw, db, None, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
-def conv2d(x, weight, bias=None, stride=1, padding=0, relu=False):
+def conv2d(x, weight, bias=None, stride=1, padding=0, relu=False,
+           bn_fuse=None):
     stride, padding = _pair(stride), _pair(padding)
     if x.is_cuda:
-        return _Conv2dFn.apply(x, weight, bias, stride, padding, relu, False)
+        return _Conv2dFn.apply(x, weight, bias, stride, padding, relu, False,
+                               bn_fuse)
     y = F.conv2d(x, weight, bias, stride, padding)
     return F.relu(y) if relu else y
 
 
-def conv2d_stats(x, weight, bias=None, stride=1, padding=0):
+def conv2d_stats(x, weight, bias=None, stride=1, padding=0, bn_fuse=None):
     """conv2d that also returns the following BatchNorm's {sum, sumsq}
     partials computed in the conv epilogue (None when unsupported: CPU,
     fp32, or the C%8!=0 stem)."""
     stride, padding = _pair(stride), _pair(padding)
     if x.is_cuda:
         y, partials = _Conv2dFn.apply(x, weight, bias, stride, padding,
-                                      False, True)
+                                      False, True, bn_fuse)
         return y, (partials if partials.numel() > 0 else None)
     return F.conv2d(x, weight, bias, stride, padding), None
 
@@ -219,7 +250,8 @@ def fork_conv2d(x, conv_module):
 class _BnAddReluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                momentum, eps, relu, residual, conv_partials=None):
+                momentum, eps, relu, residual, conv_partials=None,
+                bwd_holder=None):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         if residual is not None:
@@ -233,6 +265,10 @@ class _BnAddReluFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight, bias, mean, rstd, y)
         ctx.relu = relu
         ctx.has_residual = residual is not None
+        ctx.bwd_holder = bwd_holder
+        if bwd_holder is not None:
+            bwd_holder.mean = mean
+            bwd_holder.rstd = rstd
         return y
 
     @staticmethod
@@ -240,15 +276,24 @@ class _BnAddReluFn(torch.autograd.Function):
         ext = load_kernels_ext()
         x, weight, bias, mean, rstd, y = ctx.saved_tensors
         dy = to_channels_last(dy)
-        dx, dw, db, dres = ext.bn_relu_bwd(
-            dy, x, weight, bias, mean, rstd, y, ctx.relu, ctx.has_residual)
+        holder = ctx.bwd_holder
+        pre = holder.partials if holder is not None else None
+        if pre is not None:
+            holder.partials = None  # consume once
+            dx, dw, db, dres = ext.bn_relu_bwd(
+                dy, x, weight, bias, mean, rstd, y, ctx.relu,
+                ctx.has_residual, pre)
+        else:
+            dx, dw, db, dres = ext.bn_relu_bwd(
+                dy, x, weight, bias, mean, rstd, y, ctx.relu,
+                ctx.has_residual)
         return (dx, dw, db, None, None, None, None, None,
-                dres if ctx.has_residual else None, None)
+                dres if ctx.has_residual else None, None, None)
 
 
 def bn_add_relu(x, weight, bias, running_mean, running_var, training,
                 momentum=0.1, eps=1e-5, relu=True, residual=None,
-                conv_partials=None):
+                conv_partials=None, bwd_holder=None):
     """y = relu?(bn(x) + residual?) — single fused kernel each direction.
 
     The reference issues separate cuDNN BN and ReLU (and a separate add for
@@ -258,7 +303,7 @@ def bn_add_relu(x, weight, bias, running_mean, running_var, training,
         if training:
             return _BnAddReluFn.apply(x, weight, bias, running_mean,
                                       running_var, momentum, eps, relu,
-                                      residual, conv_partials)
+                                      residual, conv_partials, bwd_holder)
         ext = load_kernels_ext()
         return ext.bn_relu_infer(to_channels_last(x), weight, bias,
                                  running_mean, running_var, float(eps), relu,

@@ -12,7 +12,12 @@ from typing import Optional
 import torch
 import torch.nn as nn
 
+import os
+
 from . import functional as DF
+
+# A/B gate for the dgrad->BN-backward-stats fusion
+_FUSE_BNBWD = os.environ.get("DDL_FUSE_BNBWD", "1") == "1"
 
 
 class Conv2d(nn.Module):
@@ -39,13 +44,18 @@ class Conv2d(nn.Module):
         nn.init.kaiming_normal_(self.weight, mode="fan_out", nonlinearity="relu")
 
     def forward(self, x):
+        fuse = getattr(x, "_ddl_bn_bwd_fuse", None)
+        if fuse is not None:
+            del x._ddl_bn_bwd_fuse  # single consumer takes it
         if self.emit_bn_stats and self.training and x.is_cuda:
             y, partials = DF.conv2d_stats(x, self.weight, self.bias,
-                                          self.stride, self.padding)
+                                          self.stride, self.padding,
+                                          bn_fuse=fuse)
             if partials is not None:
                 y._ddl_bn_partials = partials
             return y
-        return DF.conv2d(x, self.weight, self.bias, self.stride, self.padding)
+        return DF.conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                         bn_fuse=fuse)
 
     def extra_repr(self):
         return (f"{self.in_channels}, {self.out_channels}, "
@@ -62,11 +72,18 @@ class BatchNorm2d(nn.Module):
     (fp32 statistics under bf16/fp16 activations, BASELINE.json north star).
     """
 
-    def __init__(self, num_features, eps=1e-5, momentum=0.1):
+    def __init__(self, num_features, eps=1e-5, momentum=0.1,
+                 fuse_bwd_stats=False):
         super().__init__()
         self.num_features = num_features
         self.eps = eps
         self.momentum = momentum
+        # fuse_bwd_stats: this BN's output has exactly ONE consumer — a
+        # Conv2d — so that conv's dgrad epilogue can emit this BN's
+        # backward stats partials (set by the model builder for the
+        # bn+relu layers inside a conv chain; NEVER set when the output
+        # fans out, the partials would miss the other consumers' grads)
+        self.fuse_bwd_stats = fuse_bwd_stats
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
@@ -96,10 +113,20 @@ class BatchNorm2d(nn.Module):
         partials = getattr(x, "_ddl_bn_partials", None)
         if partials is not None:
             del x._ddl_bn_partials
-        return DF.bn_add_relu(
+        holder = None
+        if (self.fuse_bwd_stats and _FUSE_BNBWD and self.training
+                and x.is_cuda and relu and residual is None
+                and x.dtype != torch.float32
+                and torch.is_grad_enabled()):
+            holder = DF.BnBwdFusion()
+        y = DF.bn_add_relu(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, relu, residual,
-            conv_partials=partials)
+            conv_partials=partials, bwd_holder=holder)
+        if holder is not None:
+            # picked up by the consuming Conv2d (read-only refs)
+            y._ddl_bn_bwd_fuse = (holder, x, self.weight, self.bias)
+        return y
 
     def extra_repr(self):
         return f"{self.num_features}, eps={self.eps}, momentum={self.momentum}"
