@@ -305,7 +305,10 @@ at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
 
 // ---- batchnorm -------------------------------------------------------------
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
+// returns (y, mean, rstd, relu_mask); relu_mask is the bit-packed [M][C/8]
+// ReLU activity, non-empty only on the training relu+residual path where
+// the backward would otherwise re-read y for the mask (mode 1 -> mode 3).
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias, at::Tensor rmean,
     at::Tensor rvar, double momentum, double eps, bool relu,
     c10::optional<at::Tensor> residual,
@@ -346,10 +349,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_relu_fwd(
   }
   const void* res = residual.has_value() && residual->defined()
                         ? residual->data_ptr() : nullptr;
-  ddl_bn_apply(x.data_ptr(), y.data_ptr(), res, mean.data_ptr<float>(),
-               rstd.data_ptr<float>(), weight.data_ptr(), bias.data_ptr(), M,
-               C, relu, dt, s);
-  return {y, mean, rstd};
+  at::Tensor mask = at::empty({0}, x.options().dtype(at::kByte));
+  uint8_t* mptr = nullptr;
+  if (relu && res != nullptr) {
+    mask = at::empty({M * (C / 8)}, x.options().dtype(at::kByte));
+    mptr = mask.data_ptr<uint8_t>();
+  }
+  ddl_bn_apply_mask(x.data_ptr(), y.data_ptr(), res, mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), weight.data_ptr(),
+                    bias.data_ptr(), M, C, relu, dt, mptr, s);
+  return {y, mean, rstd, mask};
 }
 
 at::Tensor bn_relu_infer(at::Tensor x, at::Tensor weight, at::Tensor bias,
@@ -370,15 +379,23 @@ at::Tensor bn_relu_infer(at::Tensor x, at::Tensor weight, at::Tensor bias,
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
     at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
     at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
-    bool has_residual, c10::optional<at::Tensor> pre_partials) {
+    bool has_residual, c10::optional<at::Tensor> pre_partials,
+    c10::optional<at::Tensor> relu_mask) {
   check_cl(dy, "dy"); check_cl(x, "x");
   int64_t C = x.size(1);
   int64_t M = x.numel() / C;
   auto s = cur_stream();
   int dt = dt_code(x);
-  // relu mask source: with a residual, y != bn(x) so the mask must come
-  // from y; without one, recompute (w*xhat+b > 0) and skip reading y.
-  const int relu_mode = relu ? (has_residual ? 1 : 2) : 0;
+  // relu mask source: with a residual, y != bn(x) so the mask comes from
+  // the forward's bit-packed mask (mode 3; 1/16 the bytes of re-reading
+  // y) or from y itself (mode 1); without one, recompute (w*xhat+b > 0).
+  const uint8_t* mptr = nullptr;
+  if (relu_mask.has_value() && relu_mask->numel() > 0) {
+    TORCH_CHECK(relu_mask->numel() == M * (C / 8), "relu mask size");
+    mptr = relu_mask->data_ptr<uint8_t>();
+  }
+  const int relu_mode =
+      relu ? (has_residual ? (mptr ? 3 : 1) : 2) : 0;
   at::Tensor sums;
   if (pre_partials.has_value() && pre_partials->numel() > 0) {
     // stats partials already emitted by the consuming conv's dgrad
@@ -397,11 +414,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
     int chunks = ddl_bn_stats_chunks(M, C);
     auto partials = at::empty({(int64_t)chunks * 2 * C},
                               x.options().dtype(at::kFloat));
-    ddl_bn_bwd_stats(dy.data_ptr(), x.data_ptr(), y.data_ptr(), relu_mode,
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     weight.data_ptr(), bias.data_ptr(), M, C, dt,
-                     sums.data_ptr<float>(), partials.data_ptr<float>(),
-                     chunks, s);
+    ddl_bn_bwd_stats_mask(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+                          relu_mode, mean.data_ptr<float>(),
+                          rstd.data_ptr<float>(), weight.data_ptr(),
+                          bias.data_ptr(), M, C, dt, sums.data_ptr<float>(),
+                          partials.data_ptr<float>(), chunks, mptr, s);
   }
   auto dx = at::empty_like(x);
   at::Tensor dres;
@@ -410,10 +427,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
     dres = at::empty_like(x);
     dres_p = dres.data_ptr();
   }
-  ddl_bn_bwd_apply(dy.data_ptr(), x.data_ptr(), y.data_ptr(), dx.data_ptr(),
-                   dres_p, mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                   weight.data_ptr(), bias.data_ptr(),
-                   sums.data_ptr<float>(), M, C, relu_mode, dt, s);
+  ddl_bn_bwd_apply_mask(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
+                        dx.data_ptr(), dres_p, mean.data_ptr<float>(),
+                        rstd.data_ptr<float>(), weight.data_ptr(),
+                        bias.data_ptr(), sums.data_ptr<float>(), M, C,
+                        relu_mode, dt, mptr, s);
   auto dw = at::empty_like(weight);
   auto db = at::empty_like(weight);
   ddl_bn_grads_from_sums(sums.data_ptr<float>(), dw.data_ptr(), db.data_ptr(),
@@ -592,7 +610,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
            bool has_residual) {
           return bn_relu_bwd(dy, x, weight, bias, mean, rstd, y, relu,
-                             has_residual, c10::nullopt);
+                             has_residual, c10::nullopt, c10::nullopt);
+        });
+  m.def("bn_relu_bwd",
+        [](at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
+           at::Tensor mean, at::Tensor rstd, at::Tensor y, bool relu,
+           bool has_residual, c10::optional<at::Tensor> pre_partials) {
+          return bn_relu_bwd(dy, x, weight, bias, mean, rstd, y, relu,
+                             has_residual, pre_partials, c10::nullopt);
         });
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);

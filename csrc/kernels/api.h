@@ -78,6 +78,25 @@ void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y,
                       int dtype, hipStream_t s);
 void ddl_bn_grads_from_sums(const float* sums, void* dweight, void* dbias,
                             int C, int dtype, hipStream_t s);
+// mask-aware variants: relu_mask is the bit-packed [M][C/8] ReLU activity
+// written by the forward apply (mode 3 backward reads it instead of y).
+void ddl_bn_apply_mask(const void* x, void* y, const void* residual,
+                       const float* mean, const float* rstd,
+                       const void* weight, const void* bias, int64_t M,
+                       int C, int relu, int dtype, uint8_t* relu_mask,
+                       hipStream_t s);
+void ddl_bn_bwd_stats_mask(const void* dy, const void* x, const void* y,
+                           int relu_mode, const float* mean,
+                           const float* rstd, const void* weight,
+                           const void* bias, int64_t M, int C, int dtype,
+                           float* sums, float* partials, int max_chunks,
+                           const uint8_t* relu_mask, hipStream_t s);
+void ddl_bn_bwd_apply_mask(const void* dy, const void* x, const void* y,
+                           void* dx, void* dres, const float* mean,
+                           const float* rstd, const void* weight,
+                           const void* bias, const float* sums, int64_t M,
+                           int C, int relu_mode, int dtype,
+                           const uint8_t* relu_mask, hipStream_t s);
 
 // ---- pooling (K6/K7) ------------------------------------------------------
 void ddl_maxpool_fwd(const void* x, void* y, uint8_t* idx, int N, int H,

@@ -182,13 +182,19 @@ __global__ void bn_finalize_k(const float* __restrict__ sums, int C,
 }
 
 // ---- apply: y = relu?((x-mean)*rstd*w + b + res?) -------------------------
-template <typename T, bool RELU, bool RES>
+// mask (training relu+residual path): bit-packed ReLU activity, one byte
+// per 8-channel group (MC/8 bytes = 1/16 of an activation pass) — the
+// backward then reads the mask instead of re-reading y (mode 3), saving a
+// full y pass in BOTH bn_bwd_stats and bn_bwd_apply for the residual-join
+// BNs (mode 1), the widest-channel layers of the net.
+template <typename T, bool RELU, bool RES, bool MASK = false>
 __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
                            const T* __restrict__ res,
                            const float* __restrict__ mean,
                            const float* __restrict__ rstd,
                            const T* __restrict__ weight,
-                           const T* __restrict__ bias, int64_t M, int C) {
+                           const T* __restrict__ bias, int64_t M, int C,
+                           uint8_t* __restrict__ mask = nullptr) {
   const int ncg = C / 8;
   const int64_t total = M * ncg;
   for (int64_t i = global_tid(); i < total; i += grid_stride()) {
@@ -200,15 +206,20 @@ __global__ void bn_apply_k(const T* __restrict__ x, T* __restrict__ y,
     Vec8<T> b8 = load8(bias + c0);
     Vec8<T> r8;
     if (RES) r8 = load8(res + m * C + c0);
+    uint8_t mb = 0;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float f = (to_f32(v.v[j]) - mean[c0 + j]) * rstd[c0 + j];
       f = f * to_f32(w8.v[j]) + to_f32(b8.v[j]);
       if (RES) f += to_f32(r8.v[j]);
-      if (RELU) f = fmaxf(f, 0.f);
+      if (RELU) {
+        if (MASK && f > 0.f) mb |= (uint8_t)(1u << j);
+        f = fmaxf(f, 0.f);
+      }
       v.v[j] = from_f32<T>(f);
     }
     store8(y + m * C + c0, v);
+    if (MASK) mask[m * ncg + cg] = mb;
   }
 }
 
@@ -247,7 +258,9 @@ __global__ void bn_infer_k(const T* __restrict__ x, T* __restrict__ y,
 // ---- backward stats: sums[c] = sum(dy_m * xhat), sums[C+c] = sum(dy_m) ----
 // RELU modes: 0 = no relu; 1 = mask from y (residual case — y != bn(x));
 // 2 = recompute mask as (w*xhat + b > 0), skipping the y read entirely
-// (one full activation pass saved; exact for relu-without-residual).
+// (one full activation pass saved; exact for relu-without-residual);
+// 3 = bit-packed mask saved by the forward (residual case, 1/16 the
+// bytes of the y read it replaces).
 template <typename T, int RELU>
 __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
                                const T* __restrict__ x,
@@ -256,7 +269,8 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
                                const float* __restrict__ rstd,
                                const T* __restrict__ weight,
                                const T* __restrict__ bias, int64_t M,
-                               int C, float* __restrict__ sums) {
+                               int C, float* __restrict__ sums,
+                               const uint8_t* __restrict__ mask = nullptr) {
   __shared__ float red[2][ROWS][CG * 8];
   const int vg = threadIdx.x % CG;
   const int trow = threadIdx.x / CG;
@@ -274,6 +288,8 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
     Vec8<T> vx = load8(x + m * C + c0);
     Vec8<T> vy;
     if (RELU == 1) vy = load8(y + m * C + c0);
+    uint8_t mb = 0xff;
+    if (RELU == 3) mb = mask[m * (C / 8) + c0 / 8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g = to_f32(vdy.v[j]);
@@ -281,6 +297,7 @@ __global__ void bn_bwd_stats_k(const T* __restrict__ dy,
       if (RELU == 1) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
       if (RELU == 2)
         g = (xh * to_f32(w8.v[j]) + to_f32(b8.v[j])) > 0.f ? g : 0.f;
+      if (RELU == 3) g = (mb >> j) & 1 ? g : 0.f;
       s0[j] += g * xh;
       s1[j] += g;
     }
@@ -324,7 +341,8 @@ __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
                                const T* __restrict__ weight,
                                const T* __restrict__ bias,
                                const float* __restrict__ sums, int64_t M,
-                               int C) {
+                               int C,
+                               const uint8_t* __restrict__ mask = nullptr) {
   const int ncg = C / 8;
   const int64_t total = M * ncg;
   const float invM = 1.f / (float)M;
@@ -339,6 +357,8 @@ __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
     if (RELU == 2) b8 = load8(bias + c0);
     Vec8<T> vy, out, rout;
     if (RELU == 1) vy = load8(y + m * C + c0);
+    uint8_t mb = 0xff;
+    if (RELU == 3) mb = mask[m * ncg + cg];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
@@ -347,6 +367,7 @@ __global__ void bn_bwd_apply_k(const T* __restrict__ dy,
       if (RELU == 1) g = to_f32(vy.v[j]) > 0.f ? g : 0.f;
       if (RELU == 2)
         g = (xh * to_f32(w8.v[j]) + to_f32(b8.v[j])) > 0.f ? g : 0.f;
+      if (RELU == 3) g = (mb >> j) & 1 ? g : 0.f;
       if (RES) rout.v[j] = from_f32<T>(g);
       float t = g - sums[C + c] * invM - xh * sums[c] * invM;
       out.v[j] = from_f32<T>(to_f32(w8.v[j]) * rstd[c] * t);
@@ -447,16 +468,29 @@ void ddl_bn_apply(const void* x, void* y, const void* residual,
                   const float* mean, const float* rstd, const void* weight,
                   const void* bias, int64_t M, int C, int relu, int dtype,
                   hipStream_t s) {
+  ddl_bn_apply_mask(x, y, residual, mean, rstd, weight, bias, M, C, relu,
+                    dtype, nullptr, s);
+}
+
+// relu_mask != nullptr (training relu paths): also write the bit-packed
+// ReLU mask [M][C/8] consumed by the mode-3 backward.
+void ddl_bn_apply_mask(const void* x, void* y, const void* residual,
+                       const float* mean, const float* rstd,
+                       const void* weight, const void* bias, int64_t M,
+                       int C, int relu, int dtype, uint8_t* relu_mask,
+                       hipStream_t s) {
   int g = apply_grid(M, C);
-#define RUN(T, RELU, RES)                                                      \
-  bn_apply_k<T, RELU, RES><<<g, 256, 0, s>>>(                                  \
+#define RUN(T, RELU, RES, MASK)                                                \
+  bn_apply_k<T, RELU, RES, MASK><<<g, 256, 0, s>>>(                            \
       (const T*)x, (T*)y, (const T*)residual, mean, rstd, (const T*)weight,    \
-      (const T*)bias, M, C)
+      (const T*)bias, M, C, relu_mask)
 #define SEL(T)                                                                 \
-  if (relu && residual) RUN(T, true, true);                                    \
-  else if (relu) RUN(T, true, false);                                          \
-  else if (residual) RUN(T, false, true);                                      \
-  else RUN(T, false, false)
+  if (relu && residual && relu_mask) RUN(T, true, true, true);                 \
+  else if (relu && residual) RUN(T, true, true, false);                        \
+  else if (relu && relu_mask) RUN(T, true, false, true);                       \
+  else if (relu) RUN(T, true, false, false);                                   \
+  else if (residual) RUN(T, false, true, false);                               \
+  else RUN(T, false, false, false)
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN
@@ -486,15 +520,25 @@ void ddl_bn_bwd_stats(const void* dy, const void* x, const void* y,
                       const void* weight, const void* bias, int64_t M, int C,
                       int dtype, float* sums, float* partials, int max_chunks,
                       hipStream_t s) {
+  ddl_bn_bwd_stats_mask(dy, x, y, relu_mode, mean, rstd, weight, bias, M, C,
+                        dtype, sums, partials, max_chunks, nullptr, s);
+}
+
+void ddl_bn_bwd_stats_mask(const void* dy, const void* x, const void* y,
+                           int relu_mode, const float* mean,
+                           const float* rstd, const void* weight,
+                           const void* bias, int64_t M, int C, int dtype,
+                           float* sums, float* partials, int max_chunks,
+                           const uint8_t* relu_mask, hipStream_t s) {
   dim3 g = stats_grid(M, C);
   if ((int)g.y > max_chunks) g.y = max_chunks;
 #define RUN(T, RELU)                                                           \
   bn_bwd_stats_k<T, RELU><<<g, 256, 0, s>>>(                                   \
       (const T*)dy, (const T*)x, (const T*)y, mean, rstd, (const T*)weight,    \
-      (const T*)bias, M, C, partials)
+      (const T*)bias, M, C, partials, relu_mask)
 #define SEL(T)                                                                 \
-  if (relu_mode == 2) RUN(T, 2); else if (relu_mode == 1) RUN(T, 1);           \
-  else RUN(T, 0)
+  if (relu_mode == 3) RUN(T, 3); else if (relu_mode == 2) RUN(T, 2);           \
+  else if (relu_mode == 1) RUN(T, 1); else RUN(T, 0)
   switch (dtype) { case 0: { SEL(float); } break; case 1: { SEL(bf16); } break; case 2: { SEL(f16); } break; }
 #undef SEL
 #undef RUN
@@ -506,13 +550,25 @@ void ddl_bn_bwd_apply(const void* dy, const void* x, const void* y, void* dx,
                       const void* weight, const void* bias, const float* sums,
                       int64_t M, int C, int relu_mode, int dtype,
                       hipStream_t s) {
+  ddl_bn_bwd_apply_mask(dy, x, y, dx, dres, mean, rstd, weight, bias, sums,
+                        M, C, relu_mode, dtype, nullptr, s);
+}
+
+void ddl_bn_bwd_apply_mask(const void* dy, const void* x, const void* y,
+                           void* dx, void* dres, const float* mean,
+                           const float* rstd, const void* weight,
+                           const void* bias, const float* sums, int64_t M,
+                           int C, int relu_mode, int dtype,
+                           const uint8_t* relu_mask, hipStream_t s) {
   int g = apply_grid(M, C);
 #define RUN(T, RELU, RES)                                                      \
   bn_bwd_apply_k<T, RELU, RES><<<g, 256, 0, s>>>(                              \
       (const T*)dy, (const T*)x, (const T*)y, (T*)dx, (T*)dres, mean, rstd,    \
-      (const T*)weight, (const T*)bias, sums, M, C)
+      (const T*)weight, (const T*)bias, sums, M, C, relu_mask)
 #define SEL(T)                                                                 \
-  if (relu_mode == 1 && dres) RUN(T, 1, true);                                 \
+  if (relu_mode == 3 && dres) RUN(T, 3, true);                                 \
+  else if (relu_mode == 3) RUN(T, 3, false);                                   \
+  else if (relu_mode == 1 && dres) RUN(T, 1, true);                            \
   else if (relu_mode == 2 && dres) RUN(T, 2, true);                            \
   else if (relu_mode == 2) RUN(T, 2, false);                                   \
   else if (relu_mode == 1) RUN(T, 1, false);                                   \

@@ -256,13 +256,13 @@ class _BnAddReluFn(torch.autograd.Function):
         x = to_channels_last(x)
         if residual is not None:
             residual = to_channels_last(residual)
-        y, mean, rstd = ext.bn_relu_fwd(
+        y, mean, rstd, relu_mask = ext.bn_relu_fwd(
             x, weight, bias, running_mean, running_var,
             float(momentum), float(eps), relu,
             residual if residual is not None else None,
             conv_partials,
         )
-        ctx.save_for_backward(x, weight, bias, mean, rstd, y)
+        ctx.save_for_backward(x, weight, bias, mean, rstd, y, relu_mask)
         ctx.relu = relu
         ctx.has_residual = residual is not None
         ctx.bwd_holder = bwd_holder
@@ -274,7 +274,7 @@ class _BnAddReluFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         ext = load_kernels_ext()
-        x, weight, bias, mean, rstd, y = ctx.saved_tensors
+        x, weight, bias, mean, rstd, y, relu_mask = ctx.saved_tensors
         dy = to_channels_last(dy)
         holder = ctx.bwd_holder
         pre = holder.partials if holder is not None else None
@@ -282,11 +282,11 @@ class _BnAddReluFn(torch.autograd.Function):
             holder.partials = None  # consume once
             dx, dw, db, dres = ext.bn_relu_bwd(
                 dy, x, weight, bias, mean, rstd, y, ctx.relu,
-                ctx.has_residual, pre)
+                ctx.has_residual, pre, relu_mask)
         else:
             dx, dw, db, dres = ext.bn_relu_bwd(
                 dy, x, weight, bias, mean, rstd, y, ctx.relu,
-                ctx.has_residual)
+                ctx.has_residual, None, relu_mask)
         return (dx, dw, db, None, None, None, None, None,
                 dres if ctx.has_residual else None, None, None)
 
