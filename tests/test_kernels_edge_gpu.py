@@ -81,7 +81,7 @@ def test_bn_small_spatial_deep_channels():
                               True, 0.1, 1e-5))
     from distributeddeeplearning_amd.ops._ext import load_kernels_ext
     ext = load_kernels_ext()
-    y, mean, rstd = ext.bn_relu_fwd(cl(x.to(DEV)), wgt.bfloat16().to(DEV),
+    y, mean, rstd, _mask = ext.bn_relu_fwd(cl(x.to(DEV)), wgt.bfloat16().to(DEV),
                                     b.bfloat16().to(DEV), rm.to(DEV),
                                     rv.to(DEV), 0.1, 1e-5, True, None)
     assert relerr(y, ref) < 0.02

@@ -118,7 +118,7 @@ def test_bn_fwd_train():
 
     ext = _ext()
     rm_g, rv_g = rm.clone().to(DEV), rv.clone().to(DEV)
-    y, mean, rstd = ext.bn_relu_fwd(cl(x.to(DEV)), wgt.bfloat16().to(DEV),
+    y, mean, rstd, _mask = ext.bn_relu_fwd(cl(x.to(DEV)), wgt.bfloat16().to(DEV),
                                     b.bfloat16().to(DEV), rm_g, rv_g,
                                     0.1, 1e-5, False, None)
     assert relerr(y, ref) < 0.01
@@ -141,7 +141,7 @@ def test_bn_relu_residual_fwd_bwd():
     resg = cl(res.detach().to(DEV).bfloat16())
     wgt_g = wgt.detach().bfloat16().to(DEV)
     bg = b.detach().bfloat16().to(DEV)
-    y, mean, rstd = ext.bn_relu_fwd(xg, wgt_g, bg, rm.to(DEV), rv.to(DEV),
+    y, mean, rstd, _mask = ext.bn_relu_fwd(xg, wgt_g, bg, rm.to(DEV), rv.to(DEV),
                                     0.1, 1e-5, True, resg)
     assert relerr(y, ref) < 0.02
     dyg = cl(dy.to(DEV).bfloat16())
@@ -151,6 +151,13 @@ def test_bn_relu_residual_fwd_bwd():
     assert relerr(dw, wgt.grad) < 0.02
     assert relerr(db, b.grad) < 0.02
     assert relerr(dx, x.grad) < 0.05  # bf16 stats amplify; mean-rel still small
+    # mode 3: bit-packed forward mask instead of the y re-read — must match
+    # the mode-1 (mask-from-y) results bit-for-bit
+    assert _mask.numel() == xg.numel() // 8
+    dx3, dw3, db3, dres3 = ext.bn_relu_bwd(dyg, xg, wgt_g, bg, mean, rstd, y,
+                                           True, True, None, _mask)
+    assert torch.equal(dx3, dx) and torch.equal(dres3, dres)
+    assert torch.equal(dw3, dw) and torch.equal(db3, db)
 
 
 # ---- pooling --------------------------------------------------------------
@@ -292,7 +299,7 @@ def test_bn_relu_no_residual_bwd_recomputed_mask():
     xg = cl(x.detach().to(DEV).bfloat16())
     wgt_g = wgt.detach().bfloat16().to(DEV)
     bg = b.detach().bfloat16().to(DEV)
-    y, mean, rstd = ext.bn_relu_fwd(xg, wgt_g, bg, rm.to(DEV), rv.to(DEV),
+    y, mean, rstd, _mask = ext.bn_relu_fwd(xg, wgt_g, bg, rm.to(DEV), rv.to(DEV),
                                     0.1, 1e-5, True, None)
     dx, dw, db, _ = ext.bn_relu_bwd(cl(dy.to(DEV).bfloat16()), xg, wgt_g, bg,
                                     mean, rstd, y, True, False)
@@ -318,9 +325,9 @@ def test_conv_bn_stats_fusion():
     b = torch.zeros(C, dtype=torch.bfloat16, device=DEV)
     rm0, rv0 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
     rm1, rv1 = torch.zeros(C, device=DEV), torch.ones(C, device=DEV)
-    yb0, mean0, rstd0 = ext.bn_relu_fwd(y0, wgt, b, rm0, rv0, 0.1, 1e-5,
+    yb0, mean0, rstd0, _m0 = ext.bn_relu_fwd(y0, wgt, b, rm0, rv0, 0.1, 1e-5,
                                         False, None)
-    yb1, mean1, rstd1 = ext.bn_relu_fwd(y1, wgt, b, rm1, rv1, 0.1, 1e-5,
+    yb1, mean1, rstd1, _m1 = ext.bn_relu_fwd(y1, wgt, b, rm1, rv1, 0.1, 1e-5,
                                         False, None, partials)
     # partials are pre-bf16-rounding sums: equal up to rounding noise
     assert relerr(mean1, mean0) < 5e-3
