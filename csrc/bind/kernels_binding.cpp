@@ -182,6 +182,36 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> conv_fwd_smallc(
   return {y, col, partials};
 }
 
+// stem dgrad (C % 8 != 0): dcol = dy @ wp (library GEMM, K8-class) then a
+// col2im scatter-add back to dx.  Completes K2 generality for the 7x7 C=3
+// stem (input grads are only needed when training on learned inputs).
+at::Tensor conv_dgrad_smallc(at::Tensor dy, at::Tensor w,
+                             std::vector<int64_t> x_shape, int64_t sh,
+                             int64_t sw, int64_t ph, int64_t pw) {
+  check_cl(dy, "dy"); check_cl(w, "w");
+  int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
+  int K = w.size(0), R = w.size(2), S = w.size(3);
+  int P = dy.size(2), Q = dy.size(3);
+  const int dt = dt_code(dy);
+  const int kred = R * S * C;
+  const int ld = (kred + 7) & ~7;
+  auto wp = at::empty({K, ld}, w.options());
+  ddl_pad_cols(w.data_ptr(), wp.data_ptr(), K, kred, ld, dt, cur_stream());
+  // dy rows are [N*P*Q][K] in channels_last memory
+  auto dy2d = dy.permute({0, 2, 3, 1}).reshape({(int64_t)N * P * Q, K});
+  auto dcol = at::matmul(dy2d, wp);  // [M][ld]
+  auto ws = at::zeros({(int64_t)N * H * W * C},
+                      dy.options().dtype(at::kFloat));
+  auto& tbl = im2col_table(R, S, C, ld, dy.device());
+  ddl_col2im_acc(dcol.data_ptr(), ws.data_ptr<float>(), tbl.data_ptr<int>(),
+                 N, H, W, C, P, Q, sh, sw, ph, pw, kred, ld, dt,
+                 cur_stream());
+  auto dx = empty_nhwc(N, C, H, W, dy.options());
+  ddl_cast_f32_to(ws.data_ptr<float>(), dx.data_ptr(),
+                  (int64_t)N * H * W * C, dt, cur_stream());
+  return dx;
+}
+
 at::Tensor conv_wgrad_smallc(at::Tensor dy, at::Tensor col,
                              std::vector<int64_t> w_shape,
                              std::vector<int64_t> x_shape) {
@@ -586,6 +616,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd_stats", &conv_fwd_stats);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_dgrad_bnstats", &conv_dgrad_bnstats);
+  m.def("conv_dgrad_smallc", &conv_dgrad_smallc);
   m.def("conv_dgrad",
         [](at::Tensor dy, at::Tensor w, std::vector<int64_t> xs, int64_t sh,
            int64_t sw, int64_t ph, int64_t pw) {

@@ -170,6 +170,11 @@ void ddl_conv_wgrad(const void* dy, const void* x, float* dw_ws, void* dw,
 void ddl_im2col(const void* x, void* out, const int* tbl, int N, int H,
                 int W, int C, int R, int S, int P, int Q, int sh, int sw,
                 int ph, int pw, int ldout, int dtype, hipStream_t s);
+// col2im scatter-add for the stem dgrad (fp32 atomic accumulation)
+void ddl_col2im_acc(const void* dcol, float* dx_ws, const int* tbl, int N,
+                    int H, int W, int C, int P, int Q, int sh, int sw,
+                    int ph, int pw, int kred_total, int ldout, int dtype,
+                    hipStream_t s);
 void ddl_pad_cols(const void* w, void* out, int K, int kred, int ldout,
                   int dtype, hipStream_t s);
 void ddl_narrow_cols(const void* w, void* out, int K, int kred, int ldout,

@@ -266,6 +266,36 @@ inline int ew_grid(int64_t n, int per_thread = 8, int block = 256) {
   return (int)std::min<int64_t>(blocks, 2048);
 }
 
+
+// col2im scatter-add (stem dgrad, C % 8 != 0): dcol [N*P*Q][ldout] rows are
+// the per-output-pixel im2col gradients; each (r,s,c) column element adds
+// into dx_ws[n, p*sh+r-ph, q*sw+s-pw, c] (fp32 accumulation via atomics --
+// each dx element receives ~R*S/(sh*sw) contributions from distinct rows,
+// negligible contention).  tbl packs (r<<20 | s<<10 | c) like im2col.
+template <typename T>
+__global__ void col2im_acc_k(const T* __restrict__ dcol,
+                             float* __restrict__ dx_ws,
+                             const int* __restrict__ tbl, int N, int H,
+                             int W, int C, int P, int Q, int sh, int sw,
+                             int ph, int pw, int kred_total, int ldout) {
+  const int64_t total = (int64_t)N * P * Q * kred_total;
+  for (int64_t i = global_tid(); i < total; i += grid_stride()) {
+    const int kred = (int)(i % kred_total);
+    const int64_t m = i / kred_total;
+    const float v = to_f32(dcol[m * ldout + kred]);
+    if (v == 0.f) continue;
+    const int q = (int)(m % Q);
+    const int64_t np = m / Q;
+    const int p = (int)(np % P);
+    const int n = (int)(np / P);
+    const int t = tbl[kred];
+    const int hi = p * sh + (t >> 20) - ph;
+    const int wi = q * sw + ((t >> 10) & 1023) - pw;
+    if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+      atomicAdd(dx_ws + (((int64_t)n * H + hi) * W + wi) * C + (t & 1023), v);
+  }
+}
+
 }  // namespace
 
 #define DISPATCH(dtype, F, ...)                                \
@@ -374,6 +404,20 @@ void ddl_transpose_krsc_crsk(const void* w, void* w_t, int K, int RS, int C,
                              int dtype, hipStream_t s) {
   dim3 grid(cdiv(K, 32) * cdiv(C, 32), RS);
 #define RUN(T) transpose_kc_k<T><<<grid, 256, 0, s>>>((const T*)w, (T*)w_t, K, RS, C)
+  switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
+#undef RUN
+}
+
+
+void ddl_col2im_acc(const void* dcol, float* dx_ws, const int* tbl, int N,
+                    int H, int W, int C, int P, int Q, int sh, int sw,
+                    int ph, int pw, int kred_total, int ldout, int dtype,
+                    hipStream_t s) {
+  const int64_t total = (int64_t)N * P * Q * kred_total;
+  const int g = (int)std::min<int64_t>(cdiv(total, 256), 4096);
+#define RUN(T)                                                                 \
+  col2im_acc_k<T><<<g, 256, 0, s>>>((const T*)dcol, dx_ws, tbl, N, H, W, C,   \
+                                    P, Q, sh, sw, ph, pw, kred_total, ldout)
   switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
 }

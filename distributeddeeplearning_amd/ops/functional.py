@@ -122,21 +122,27 @@ class _Conv2dFn(torch.autograd.Function):
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
             if ctx.small_c:
-                raise RuntimeError("dgrad for C%8!=0 (stem) is not needed "
-                                   "(input has no grad) and not implemented")
-            fuse = ctx.bn_fuse
-            if (fuse is not None and fuse[0].mean is not None
-                    and x.dtype != torch.float32):
-                holder, bnx, bn_w, bn_b = fuse
-                dx, bparts = ext.conv_dgrad_bnstats(
-                    dy, weight, list(x.shape), ctx.stride[0], ctx.stride[1],
-                    ctx.padding[0], ctx.padding[1], to_channels_last(bnx),
-                    holder.mean, holder.rstd, bn_w, bn_b)
-                holder.partials = bparts
+                # stem path: dcol = dy @ w_pad (library GEMM) + col2im
+                # scatter-add — completes K2 for C % 8 != 0 inputs
+                dx = ext.conv_dgrad_smallc(dy, weight, list(x.shape),
+                                           ctx.stride[0], ctx.stride[1],
+                                           ctx.padding[0], ctx.padding[1])
+                fuse = None
             else:
-                dx = ext.conv_dgrad(dy, weight, list(x.shape),
-                                    ctx.stride[0], ctx.stride[1],
-                                    ctx.padding[0], ctx.padding[1])
+                fuse = ctx.bn_fuse
+                if (fuse is not None and fuse[0].mean is not None
+                        and x.dtype != torch.float32):
+                    holder, bnx, bn_w, bn_b = fuse
+                    dx, bparts = ext.conv_dgrad_bnstats(
+                        dy, weight, list(x.shape), ctx.stride[0],
+                        ctx.stride[1], ctx.padding[0], ctx.padding[1],
+                        to_channels_last(bnx), holder.mean, holder.rstd,
+                        bn_w, bn_b)
+                    holder.partials = bparts
+                else:
+                    dx = ext.conv_dgrad(dy, weight, list(x.shape),
+                                        ctx.stride[0], ctx.stride[1],
+                                        ctx.padding[0], ctx.padding[1])
         if ctx.needs_input_grad[1]:
             if ctx.small_c:
                 dw = ext.conv_wgrad_smallc(dy, col, list(weight.shape),

@@ -217,3 +217,24 @@ def test_fused_sgd_two_param_groups():
         opt.step()
         opt_c.step()
     assert relerr(a, a_c) < 1e-6 and relerr(b, b_c) < 1e-6
+
+
+def test_stem_dgrad_smallc():
+    """K2 completeness: dgrad through the C=3 stem (im2col GEMM + col2im
+    scatter-add) vs the plain-PyTorch reference."""
+    import torch.nn.functional as F
+    from distributeddeeplearning_amd.ops import functional as DF
+    torch.manual_seed(77)
+    x = torch.randn(4, 3, 34, 34).float().requires_grad_(True)
+    w = (torch.randn(64, 3, 7, 7) * 0.1).float().requires_grad_(True)
+    ref = F.conv2d(x, w, stride=2, padding=3)
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+
+    xg = cl(x.detach().to(DEV).bfloat16()).requires_grad_(True)
+    wg = cl(w.detach().to(DEV).bfloat16()).requires_grad_(True)
+    y = DF.conv2d(xg, wg, stride=2, padding=3)
+    y.backward(cl(dy.to(DEV).bfloat16()))
+    assert relerr(y, ref) < 0.02
+    assert relerr(wg.grad, w.grad) < 0.03
+    assert relerr(xg.grad, x.grad) < 0.03
