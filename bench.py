@@ -70,6 +70,7 @@ def main() -> None:
             head = None  # gpurun snapshots exclude .git
         provenance = {
             "comm_backend": api.comm_backend_name(),
+            "hipgraph": bool(getattr(trainer, "used_hipgraph", False)),
             "bucket_mb": cfg.bucket_mb,
             "n_buckets": len(trainer.opt.buckets.buckets),
             "git_head": head,
