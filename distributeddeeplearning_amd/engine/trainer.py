@@ -243,8 +243,14 @@ class Trainer:
     def _can_graph(self) -> bool:
         """hipGraph-capture eligibility for the bench loop: single GPU (RCCL
         collectives are not captured), no dynamic loss scaling (host-side
-        control flow) and no dropout (philox seed would freeze per graph)."""
-        return (os.environ.get("DDL_HIPGRAPH", "1") == "1"
+        control flow) and no dropout (philox seed would freeze per graph).
+
+        Measured (gpurun, b64/128/256/1024): replay is NEUTRAL for this
+        training step — the big MFMA/BN kernels average tens to hundreds
+        of us, so eager launches are fully hidden; opt-in via
+        DDL_HIPGRAPH=1 (it pays in launch-bound regimes like small-batch
+        inference — tools/infer_bench.py)."""
+        return (os.environ.get("DDL_HIPGRAPH", "0") == "1"
                 and self.device.type == "cuda" and self.world == 1
                 and not self.scaler.enabled
                 and self.cfg.model.startswith("resnet"))

@@ -21,6 +21,8 @@ def _cfg(**kw):
 
 
 def test_graph_capture_replays_train():
+    import os
+    os.environ["DDL_HIPGRAPH"] = "1"
     from distributeddeeplearning_amd.engine.trainer import Trainer
     tr = Trainer(_cfg(), steps_per_epoch=10 ** 9)
     assert tr._can_graph()
@@ -44,6 +46,8 @@ def test_graph_capture_replays_train():
 def test_bench_uses_graphs_and_stays_sane():
     from distributeddeeplearning_amd.engine.trainer import Trainer
     tr = Trainer(_cfg(), steps_per_epoch=10 ** 9)
+    import os
+    os.environ["DDL_HIPGRAPH"] = "1"
     res = tr.bench(steps=6, warmup=2)
     assert tr.used_hipgraph is True
     assert res["images_per_sec"] > 0
