@@ -1,0 +1,85 @@
+#!/usr/bin/env python3
+"""Batch-8k fp16+LARS convergence validation (BASELINE config 3 acceptance,
+VERDICT r1 #5).
+
+Runs the global-batch-8192 recipe on one GPU (batch 1024 x 8 accumulation
+passes — the Horovod-contract N-backwards-then-step pattern, numerically
+the DP=8 trajectory since grads average identically) and a matching fp32
+reference, and writes both loss curves + the dynamic-loss-scale event log
+to JSON for profiles/lars_b8k_convergence.md.
+
+    python tools/lars_b8k.py --steps 100 --dtype fp16  --out gpurun_out/lars_fp16.json
+    python tools/lars_b8k.py --steps 100 --dtype fp32 --out gpurun_out/lars_fp32.json
+"""
+import argparse
+import dataclasses
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=100, help="optimizer steps")
+    p.add_argument("--dtype", default="fp16")
+    p.add_argument("--batch", type=int, default=1024)
+    p.add_argument("--accum", type=int, default=8)
+    p.add_argument("--out", default="gpurun_out/lars_curve.json")
+    args = p.parse_args()
+
+    import torch
+    from distributeddeeplearning_amd.config import PRESETS
+    from distributeddeeplearning_amd.engine.trainer import Trainer
+
+    cfg = dataclasses.replace(
+        PRESETS["resnet50-fp16-lars-b8k"],
+        dtype=args.dtype, batch_size=args.batch,
+        backward_passes_per_step=args.accum, comm="single",
+    )
+    # steps_per_epoch from the GLOBAL batch (8192): the 5-epoch warmup
+    # covers ~780 steps, so a few hundred steps sit inside warmup exactly
+    # as the large-batch recipe intends (Goyal et al.)
+    spe = max(1, 1_281_167 // (args.batch * args.accum))
+    tr = Trainer(cfg, steps_per_epoch=spe)
+    # correct the LR schedule for the accumulated global batch (world=1)
+    from distributeddeeplearning_amd.engine.schedule import WarmupPiecewiseLR
+    tr.lr_sched = WarmupPiecewiseLR(
+        tr.opt, cfg.base_lr, args.batch * args.accum, spe,
+        cfg.warmup_epochs, cfg.lr_decay_epochs, cfg.lr_decay_factor)
+
+    it = tr.data_iter()
+    curve = []
+    scale_events = []
+    last_scale = tr.scaler.scale
+    for step in range(args.steps):
+        losses = []
+        for _ in range(args.accum):
+            x, y = next(it)
+            loss = tr.train_step(x, y)
+            losses.append(float(loss.detach().float()))
+        if tr.scaler.scale != last_scale:
+            scale_events.append({"step": step, "from": last_scale,
+                                 "to": tr.scaler.scale})
+            last_scale = tr.scaler.scale
+        curve.append(sum(losses) / len(losses))
+        if step % 10 == 0:
+            print(f"step {step:4d} loss {curve[-1]:.4f} lr "
+                  f"{tr.lr_sched.last_lr:.4f} scale {tr.scaler.scale:.0f}",
+                  flush=True)
+    out = {
+        "dtype": args.dtype, "global_batch": args.batch * args.accum,
+        "steps": args.steps, "loss": curve,
+        "loss_scale_events": scale_events,
+        "skipped_steps": tr.scaler.skipped_steps,
+        "final_scale": tr.scaler.scale,
+    }
+    os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
+    with open(args.out, "w") as f:
+        json.dump(out, f)
+    print("wrote", args.out)
+
+
+if __name__ == "__main__":
+    main()
