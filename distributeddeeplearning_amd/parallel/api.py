@@ -122,23 +122,54 @@ def barrier() -> None:
 
 # ---- fused state broadcast (C2/C3 call sites) ----------------------------
 
+_FUSED_BCAST_CHUNK = 64 << 20  # bytes; VGG-16 fp32 would otherwise stage 553 MB
+
+
 def _fused_broadcast(tensors: Iterable[torch.Tensor], root_rank: int) -> None:
-    """Broadcast many tensors as one flat buffer per (device, dtype) group."""
+    """Broadcast many tensors as flat buffers per (device, dtype) group.
+
+    Chunked to ~64 MB staging buffers so startup never materialises a full
+    model copy; each chunk is one collective, still
+    orders of magnitude fewer than the reference's per-tensor broadcasts."""
     groups: dict[tuple, list[torch.Tensor]] = {}
     for t in tensors:
         if not torch.is_tensor(t) or t.numel() == 0:
             continue
         groups.setdefault((t.device, t.dtype), []).append(t)
-    for (_dev, _dtype), ts in groups.items():
-        flat = torch.cat([t.detach().reshape(-1) for t in ts])
-        broadcast_(flat, root_rank)
-        off = 0
+    for (_dev, dtype), ts in groups.items():
+        max_el = max(1, _FUSED_BCAST_CHUNK // max(1, ts[0].element_size()))
+        chunk: list[torch.Tensor] = []
+        n_el = 0
+
+        def flush():
+            nonlocal chunk, n_el
+            if not chunk:
+                return
+            if len(chunk) == 1 and chunk[0].detach().reshape(-1).data_ptr() \
+                    == chunk[0].data_ptr() and chunk[0].is_contiguous():
+                # single contiguous tensor: broadcast in place, no staging
+                broadcast_(chunk[0].detach().reshape(-1), root_rank)
+            else:
+                flat = torch.cat([t.detach().reshape(-1) for t in chunk])
+                broadcast_(flat, root_rank)
+                off = 0
+                for t in chunk:
+                    n = t.numel()
+                    # NB: copy_ INTO the tensor (reshape(-1) on a
+                    # channels_last tensor returns a copy, so writing
+                    # through it would be lost)
+                    t.detach().copy_(flat[off : off + n].view(t.shape))
+                    off += n
+            chunk, n_el = [], 0
+
         for t in ts:
-            n = t.numel()
-            # NB: copy_ INTO the tensor (reshape(-1) on a channels_last
-            # tensor returns a copy, so writing through it would be lost)
-            t.detach().copy_(flat[off : off + n].view(t.shape))
-            off += n
+            if n_el and n_el + t.numel() > max_el:
+                flush()
+            chunk.append(t)
+            n_el += t.numel()
+            if t.numel() >= max_el:
+                flush()
+        flush()
 
 
 def broadcast_parameters(params, root_rank: int = 0) -> None:
