@@ -300,15 +300,27 @@ std::tuple<at::Tensor, at::Tensor> conv_dgrad_bnstats(
   return {dx, partials};
 }
 
+// out (optional): write dw into this channels_last tensor (the flat-bucket
+// grad view — eliminates the AccumulateGrad add) instead of allocating.
 at::Tensor conv_wgrad(at::Tensor dy, at::Tensor x,
                       std::vector<int64_t> w_shape, int64_t sh, int64_t sw,
-                      int64_t ph, int64_t pw) {
+                      int64_t ph, int64_t pw,
+                      c10::optional<at::Tensor> out = c10::nullopt) {
   check_cl(dy, "dy"); check_cl(x, "x");
   int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
   int K = w_shape[0], R = w_shape[2], S = w_shape[3];
   int P = dy.size(2), Q = dy.size(3);
   int dt = dt_code(x);
-  auto dw = empty_nhwc(K, C, R, S, x.options());
+  at::Tensor dw;
+  if (out.has_value() && out->defined()) {
+    check_cl(*out, "out");
+    TORCH_CHECK(out->sizes() == at::IntArrayRef(w_shape) &&
+                    out->scalar_type() == x.scalar_type(),
+                "wgrad out tensor shape/dtype mismatch");
+    dw = *out;
+  } else {
+    dw = empty_nhwc(K, C, R, S, x.options());
+  }
   if (C % (dt == 0 ? 4 : 8) != 0) {
     const int kred = R * S * C;
     const int ld = (kred + 7) & ~7;
@@ -623,6 +635,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           return conv_dgrad(dy, w, xs, sh, sw, ph, pw, c10::nullopt);
         });
   m.def("conv_wgrad", &conv_wgrad);
+  m.def("conv_wgrad",
+        [](at::Tensor dy, at::Tensor x, std::vector<int64_t> ws, int64_t sh,
+           int64_t sw, int64_t ph, int64_t pw) {
+          return conv_wgrad(dy, x, ws, sh, sw, ph, pw, c10::nullopt);
+        });
   m.def("conv_fwd_smallc", &conv_fwd_smallc);
   m.def("conv_wgrad_smallc", &conv_wgrad_smallc);
   m.def("bn_relu_fwd", &bn_relu_fwd);

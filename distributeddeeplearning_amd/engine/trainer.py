@@ -129,7 +129,11 @@ class Trainer:
         ``backward_passes_per_step``-th call also reduces and applies the
         update (the Horovod accumulation contract: N backwards, one step)."""
         tl = self.timeline
-        lr = self.lr_sched.step(self.global_step)
+        # the schedule counts OPTIMIZER steps: with gradient accumulation a
+        # window of N micro-batches advances the LR once (Goyal warmup is
+        # defined over global-batch steps)
+        lr = self.lr_sched.step(
+            self.global_step // self.cfg.backward_passes_per_step)
         if tl:
             tl.begin("forward")
             tl.begin_device("forward")

@@ -25,6 +25,41 @@ from .bucket import BucketManager
 from .compression import Compression
 
 STALL_WARN_SECONDS = float(os.environ.get("DDL_STALL_WARN_SECONDS", "60"))
+_DIRECT_WGRAD = os.environ.get("DDL_DIRECT_WGRAD", "1") == "1"
+
+
+class _DirectGradRegistry:
+    """Routes conv weight gradients straight into their flat-bucket views
+    (ops/functional.py consults this from the conv backward): `claim`
+    hands out the view only on the FIRST backward pass of an accumulation
+    window (later passes must take the autograd-accumulate path so the
+    running sum is preserved); `commit` replays exactly what the
+    post-accumulate-grad hook would have done — stall timestamp, delay
+    decrement, bucket-ready bookkeeping, collective launch."""
+
+    def __init__(self, opt: "DistributedOptimizer"):
+        self.opt = opt
+        self._members = {id(p) for b in opt.buckets.buckets for p in b.params}
+
+    def claim(self, p):
+        opt = self.opt
+        d = opt._delay.get(id(p))
+        if (id(p) not in self._members or d is None
+                or d != opt.backward_passes_per_step):
+            return None
+        g = p.grad
+        if g is None or g.data_ptr() != opt.buckets.ptr_of(p):
+            return None
+        return g
+
+    def commit(self, p):
+        opt = self.opt
+        opt._last_ready_time = time.monotonic()
+        opt._delay[id(p)] -= 1
+        if opt._delay[id(p)] == 0:
+            bucket = opt.buckets.mark_ready(p)
+            if bucket is not None:
+                opt._launch(bucket)
 
 
 class DistributedOptimizer:
@@ -80,6 +115,9 @@ class DistributedOptimizer:
                 self._delay[id(p)] = self.backward_passes_per_step
                 h = p.register_post_accumulate_grad_hook(self._make_hook())
                 self._hooks.append(h)
+        if _DIRECT_WGRAD:
+            from ..ops import functional as DF
+            DF.set_direct_wgrad(_DirectGradRegistry(self))
 
     # -- hook & launch ------------------------------------------------------
 
