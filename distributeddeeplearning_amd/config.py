@@ -90,7 +90,14 @@ class TrainConfig:
                 parser.add_argument(f"--{name}", type=lambda s: tuple(int(x) for x in s.split(",")),
                                     default=default)
             else:
-                typ = type(default) if default is not None else str
+                if default is not None:
+                    typ = type(default)
+                else:
+                    # Optional[...] fields: infer from the annotation so
+                    # e.g. --steps parses as int, not str
+                    ann = str(f.type)
+                    typ = int if "int" in ann else (
+                        float if "float" in ann else str)
                 parser.add_argument(f"--{name}", type=typ, default=default)
         parser.add_argument("--preset", type=str, default=None,
                             help=f"named preset: {sorted(PRESETS)}")
