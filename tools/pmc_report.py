@@ -15,7 +15,10 @@ with open(src) as f:
     ccol = next(c for c in r.fieldnames if "Counter_Name" in c)
     vcol = next(c for c in r.fieldnames if "Counter_Value" in c)
     for row in r:
-        k = row[kcol].split("(")[0][:56]
+        k = row[kcol]
+        k = k.replace("void (anonymous namespace)::", "")
+        k = k.replace("(anonymous namespace)::", "")
+        k = k.split("(")[0][:56]
         try:
             vals[k][row[ccol]].append(float(row[vcol]))
         except ValueError:
