@@ -268,7 +268,11 @@ def create_backend(comm: Optional[str] = None) -> CommBackend:
     MASTER_PORT; the reference used mpirun env — Horovod
     ``horovod/runner``), with gloo as the bootstrap process group.
     """
-    comm = comm or os.environ.get("DDL_COMM", "auto")
+    # DDL_COMM env wins over an explicit "auto" so the launcher/driver can
+    # force the backend (e.g. DDL_COMM=native makes a native-init failure
+    # fatal instead of falling back)
+    if comm in (None, "auto"):
+        comm = os.environ.get("DDL_COMM", "auto")
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     if world_size <= 1 and "RANK" not in os.environ:
         return SingleProcessBackend()

@@ -215,3 +215,19 @@ def _w_eval_metrics(rank, world):
 def test_average_metric_world2():
     res = run_distributed(_w_eval_metrics, world=2)
     assert all(abs(r["avg"] - 0.5) < 1e-12 for r in res)
+
+
+def _w_ddl_comm_env(rank, world):
+    import os
+    os.environ["DDL_COMM"] = "torchdist"
+    from distributeddeeplearning_amd.parallel import api
+    api.init("auto")  # explicit "auto" must still honour DDL_COMM
+    name = api.comm_backend_name()
+    api.shutdown()
+    os.environ.pop("DDL_COMM", None)
+    return {"backend": name}
+
+
+def test_ddl_comm_env_overrides_auto():
+    res = run_distributed(_w_ddl_comm_env, world=2)
+    assert all(r["backend"] == "torchdist" for r in res)
