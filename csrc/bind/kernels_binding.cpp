@@ -270,7 +270,8 @@ at::Tensor conv_dgrad(at::Tensor dy, at::Tensor w,
 std::tuple<at::Tensor, at::Tensor> conv_dgrad_bnstats(
     at::Tensor dy, at::Tensor w, std::vector<int64_t> x_shape, int64_t sh,
     int64_t sw, int64_t ph, int64_t pw, at::Tensor bnx, at::Tensor bn_mean,
-    at::Tensor bn_rstd, at::Tensor bn_w, at::Tensor bn_b) {
+    at::Tensor bn_rstd, at::Tensor bn_w, at::Tensor bn_b,
+    c10::optional<at::Tensor> accum, c10::optional<at::Tensor> bn_mask) {
   check_cl(dy, "dy"); check_cl(w, "w"); check_cl(bnx, "bnx");
   int N = x_shape[0], C = x_shape[1], H = x_shape[2], W = x_shape[3];
   int K = w.size(0), R = w.size(2), S = w.size(3);
@@ -285,6 +286,20 @@ std::tuple<at::Tensor, at::Tensor> conv_dgrad_bnstats(
               bn_rstd.scalar_type() == at::kFloat);
   TORCH_CHECK(bn_w.scalar_type() == dy.scalar_type() &&
               bn_b.scalar_type() == dy.scalar_type());
+  const void* aptr = nullptr;
+  if (accum.has_value() && accum->defined()) {
+    check_cl(*accum, "accum");
+    TORCH_CHECK(accum->scalar_type() == dy.scalar_type() &&
+                    accum->numel() == (int64_t)N * C * H * W,
+                "bnstats accum shape/dtype mismatch");
+    aptr = accum->data_ptr();
+  }
+  const uint8_t* mptr = nullptr;
+  if (bn_mask.has_value() && bn_mask->defined() && bn_mask->numel() > 0) {
+    TORCH_CHECK(bn_mask->numel() == (int64_t)N * H * W * (C / 8),
+                "bnstats mask size mismatch");
+    mptr = bn_mask->data_ptr<uint8_t>();
+  }
   auto w_t = at::empty({(int64_t)C * R * S * K}, w.options());
   ddl_transpose_krsc_crsk(w.data_ptr(), w_t.data_ptr(), K, R * S, C,
                           dt_code(w), cur_stream());
@@ -294,9 +309,9 @@ std::tuple<at::Tensor, at::Tensor> conv_dgrad_bnstats(
   ddl_conv_dgrad_bnstats(dy.data_ptr(), w_t.data_ptr(), dx.data_ptr(),
                          bnx.data_ptr(), bn_mean.data_ptr<float>(),
                          bn_rstd.data_ptr<float>(), bn_w.data_ptr(),
-                         bn_b.data_ptr(), partials.data_ptr<float>(), N, H,
-                         W, C, K, R, S, P, Q, sh, sw, ph, pw, dt,
-                         cur_stream());
+                         bn_b.data_ptr(), partials.data_ptr<float>(), aptr,
+                         mptr, N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw,
+                         dt, cur_stream());
   return {dx, partials};
 }
 
@@ -442,8 +457,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_relu_bwd(
   if (pre_partials.has_value() && pre_partials->numel() > 0) {
     // stats partials already emitted by the consuming conv's dgrad
     // epilogue (conv_dgrad_bnstats): reduce them, skip the stats pass
-    TORCH_CHECK(relu_mode == 2, "pre-computed bwd partials imply the "
-                "recomputed-mask (no-residual) path");
+    TORCH_CHECK(relu_mode == 2 || relu_mode == 3,
+                "pre-computed bwd partials need a recomputable (mode 2) or "
+                "saved-mask (mode 3) ReLU");
     const int64_t pchunks = pre_partials->numel() / (2 * C);
     TORCH_CHECK(pchunks * 2 * C == pre_partials->numel(),
                 "bwd partials shape mismatch");
@@ -628,6 +644,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd_stats", &conv_fwd_stats);
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_dgrad_bnstats", &conv_dgrad_bnstats);
+  m.def("conv_dgrad_bnstats",
+        [](at::Tensor dy, at::Tensor w, std::vector<int64_t> xs, int64_t sh,
+           int64_t sw, int64_t ph, int64_t pw, at::Tensor bnx,
+           at::Tensor bn_mean, at::Tensor bn_rstd, at::Tensor bn_w,
+           at::Tensor bn_b) {
+          return conv_dgrad_bnstats(dy, w, xs, sh, sw, ph, pw, bnx, bn_mean,
+                                    bn_rstd, bn_w, bn_b, c10::nullopt,
+                                    c10::nullopt);
+        });
   m.def("conv_dgrad_smallc", &conv_dgrad_smallc);
   m.def("conv_dgrad",
         [](at::Tensor dy, at::Tensor w, std::vector<int64_t> xs, int64_t sh,

@@ -147,13 +147,17 @@ void ddl_conv_dgrad(const void* dy, const void* w_t, void* dx, int N, int H,
 // skips its bn_bwd_stats pass.
 int64_t ddl_conv_dgrad_bnstats_chunks(int N, int H, int W, int C, int K,
                                       int sh, int sw);
+// accum: optional residual-join gradient added in the epilogue BEFORE the
+// stats (the fork-conv case); bn_mask: optional bit-packed ReLU mask from
+// the BN's forward (residual-join BNs) replacing the recomputed mask.
 void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
                             const void* bnx, const float* bn_mean,
                             const float* bn_rstd, const void* bn_w,
-                            const void* bn_b, float* bstats_ws, int N, int H,
-                            int W, int C, int K, int R, int S, int P, int Q,
-                            int sh, int sw, int ph, int pw, int dtype,
-                            hipStream_t s);
+                            const void* bn_b, float* bstats_ws,
+                            const void* accum, const uint8_t* bn_mask,
+                            int N, int H, int W, int C, int K, int R, int S,
+                            int P, int Q, int sh, int sw, int ph, int pw,
+                            int dtype, hipStream_t s);
 // reduce bwd partials [chunks][2][C] -> sums[2][C] (deterministic)
 void ddl_bn_reduce_sums(const float* partials, float* fold_ws, int chunks,
                         int C, hipStream_t s);

@@ -70,7 +70,8 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
     const float* __restrict__ bn_rstd = nullptr,
     const T* __restrict__ bn_w = nullptr,
     const T* __restrict__ bn_b = nullptr,
-    float* __restrict__ bstats_ws = nullptr, int bchunk_base = 0) {
+    float* __restrict__ bstats_ws = nullptr, int bchunk_base = 0,
+    const uint8_t* __restrict__ bn_mask = nullptr) {
   constexpr int AROWS = TBM / 32;     // staged A rows per thread
   constexpr int BROWS = TBN / 32;     // staged B rows per thread
   constexpr int ATB = TBM * 128;      // A tile bytes (128 B rows)
@@ -448,8 +449,10 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
       const bool ok = c < Nc;
       bmean[j] = ok ? bn_mean[c] : 0.f;
       brstd[j] = ok ? bn_rstd[c] : 0.f;
-      bw8[j] = ok ? to_f32(bn_w[c]) : 0.f;
-      bb8[j] = ok ? to_f32(bn_b[c]) : 0.f;
+      // with a saved bit-mask (residual-join BNs, bn.hip mode 3) the
+      // ReLU activity comes from the mask, not from w*xhat+b
+      bw8[j] = (ok && bn_mask == nullptr) ? to_f32(bn_w[c]) : 0.f;
+      bb8[j] = (ok && bn_mask == nullptr) ? to_f32(bn_b[c]) : 0.f;
     }
   }
 #pragma unroll
@@ -472,14 +475,19 @@ __global__ __launch_bounds__(256) void conv_igemm_k(
           ue[j] = from_f32<T>(to_f32(ue[j]) + to_f32(ae[j]));
       }
       if constexpr (BSTATS) {
-        // g = this dgrad row (the BN's dy), masked by the recomputed ReLU
+        // g = this dgrad row (the BN's dy, residual grad already
+        // accumulated above), masked by the recomputed or saved ReLU
         uint4 bx = *reinterpret_cast<const uint4*>(bnx + off);
         const T* be = reinterpret_cast<const T*>(&bx);
         const T* ue = reinterpret_cast<const T*>(&u);
+        uint8_t mb = 0;
+        if (bn_mask != nullptr) mb = bn_mask[off >> 3];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float xh = (to_f32(be[j]) - bmean[j]) * brstd[j];
-          if (xh * bw8[j] + bb8[j] > 0.f) {
+          const bool on = bn_mask != nullptr ? ((mb >> j) & 1)
+                                             : (xh * bw8[j] + bb8[j] > 0.f);
+          if (on) {
             const float g = to_f32(ue[j]);
             bs0[j] += g * xh;
             bs1[j] += g;
@@ -755,10 +763,11 @@ int64_t ddl_conv_dgrad_bnstats_chunks(int N, int H, int W, int C, int K,
 void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
                             const void* bnx, const float* bn_mean,
                             const float* bn_rstd, const void* bn_w,
-                            const void* bn_b, float* bstats_ws, int N, int H,
-                            int W, int C, int K, int R, int S, int P, int Q,
-                            int sh, int sw, int ph, int pw, int dtype,
-                            hipStream_t s) {
+                            const void* bn_b, float* bstats_ws,
+                            const void* accum, const uint8_t* bn_mask,
+                            int N, int H, int W, int C, int K, int R, int S,
+                            int P, int Q, int sh, int sw, int ph, int pw,
+                            int dtype, hipStream_t s) {
   ConvDims d{N, H, W, C, K, R, S, P, Q, sh, sw, ph, pw};
   const int Nc = C;
   if (sh == 1 && sw == 1) {

@@ -44,7 +44,7 @@ class BasicBlock(nn.Module):
         # stats come from conv2's dgrad epilogue (BnBwdFusion)
         self.bn1 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = BatchNorm2d(planes)
+        self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=True)  # join BN (BasicBlock)
         self.downsample = downsample
         self.stride = stride
 
@@ -72,7 +72,11 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(planes, planes, stride)
         self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = BatchNorm2d(planes * self.expansion)
+        # bn3's join output has ONE autograd consumer too (the next
+        # block's fork-conv, which carries the identity grad through its
+        # accum epilogue) — its backward stats fuse there, masked by the
+        # forward's bit-packed ReLU mask
+        self.bn3 = BatchNorm2d(planes * self.expansion, fuse_bwd_stats=True)
         self.downsample = downsample
         self.stride = stride
 

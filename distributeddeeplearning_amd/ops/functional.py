@@ -61,12 +61,13 @@ class BnBwdFusion:
     the conv's backward fills partials; the BN's backward (which autograd
     runs after the conv's) consumes them."""
 
-    __slots__ = ("mean", "rstd", "partials")
+    __slots__ = ("mean", "rstd", "partials", "mask")
 
     def __init__(self):
         self.mean = None
         self.rstd = None
         self.partials = None
+        self.mask = None  # bit-packed ReLU mask (residual-join BNs)
 
 
 class _Conv2dFn(torch.autograd.Function):
@@ -152,7 +153,7 @@ class _Conv2dFn(torch.autograd.Function):
                         dy, weight, list(x.shape), ctx.stride[0],
                         ctx.stride[1], ctx.padding[0], ctx.padding[1],
                         to_channels_last(bnx), holder.mean, holder.rstd,
-                        bn_w, bn_b)
+                        bn_w, bn_b, None, holder.mask)
                     holder.partials = bparts
                 else:
                     dx = ext.conv_dgrad(dy, weight, list(x.shape),
@@ -209,7 +210,7 @@ class _ForkConvFn(torch.autograd.Function):
     at::native elementwise cost)."""
 
     @staticmethod
-    def forward(ctx, x, weight, stride, padding, emit_stats):
+    def forward(ctx, x, weight, stride, padding, emit_stats, bn_fuse=None):
         ext = load_kernels_ext()
         x = to_channels_last(x)
         weight = to_channels_last(weight)
@@ -224,6 +225,7 @@ class _ForkConvFn(torch.autograd.Function):
         ctx.save_for_backward(x, weight)
         ctx.stride = stride
         ctx.padding = padding
+        ctx.bn_fuse = bn_fuse
         ident = x.view_as(x)
         ctx.mark_non_differentiable(partials)
         return ident, y, partials
@@ -236,9 +238,23 @@ class _ForkConvFn(torch.autograd.Function):
         dx = dw = None
         if ctx.needs_input_grad[0]:
             acc = to_channels_last(dident) if dident is not None else None
-            dx = ext.conv_dgrad(dy, weight, list(x.shape), ctx.stride[0],
-                                ctx.stride[1], ctx.padding[0],
-                                ctx.padding[1], acc)
+            fuse = ctx.bn_fuse
+            if (fuse is not None and fuse[0].mean is not None
+                    and fuse[0].mask is not None
+                    and x.dtype != torch.float32):
+                # residual-join BN backward-stats fusion: the accumulated
+                # dgrad output IS the bn3 dy; the saved bit-mask supplies
+                # the ReLU activity (bn.hip mode 3)
+                holder, bnx, bn_w, bn_b = fuse
+                dx, bparts = ext.conv_dgrad_bnstats(
+                    dy, weight, list(x.shape), ctx.stride[0], ctx.stride[1],
+                    ctx.padding[0], ctx.padding[1], to_channels_last(bnx),
+                    holder.mean, holder.rstd, bn_w, bn_b, acc, holder.mask)
+                holder.partials = bparts
+            else:
+                dx = ext.conv_dgrad(dy, weight, list(x.shape), ctx.stride[0],
+                                    ctx.stride[1], ctx.padding[0],
+                                    ctx.padding[1], acc)
         if ctx.needs_input_grad[1]:
             dst = _grad_direct.claim(weight) if _grad_direct else None
             dw = ext.conv_wgrad(dy, x, list(weight.shape), ctx.stride[0],
@@ -262,11 +278,14 @@ def fork_conv2d(x, conv_module):
                  and x.shape[1] % (4 if x.dtype == torch.float32 else 8) == 0)
     if not run_fused:
         return x, conv_module(x)
+    fuse = getattr(x, "_ddl_bn_bwd_fuse", None)
+    if fuse is not None:
+        del x._ddl_bn_bwd_fuse  # this fork is the single autograd consumer
     emit = (conv_module.emit_bn_stats and conv_module.training
             and x.dtype != torch.float32)
     ident, y, partials = _ForkConvFn.apply(
         x, conv_module.weight, conv_module.stride, conv_module.padding,
-        emit)
+        emit, fuse)
     if partials.numel() > 0:
         y._ddl_bn_partials = partials
     return ident, y
@@ -298,6 +317,7 @@ class _BnAddReluFn(torch.autograd.Function):
         if bwd_holder is not None:
             bwd_holder.mean = mean
             bwd_holder.rstd = rstd
+            bwd_holder.mask = relu_mask if relu_mask.numel() > 0 else None
         return y
 
     @staticmethod

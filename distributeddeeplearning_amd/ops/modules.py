@@ -114,8 +114,11 @@ class BatchNorm2d(nn.Module):
         if partials is not None:
             del x._ddl_bn_partials
         holder = None
+        # no-residual BNs: the consumer conv recomputes the ReLU mask;
+        # residual-join BNs: the forward's bit-packed mask rides along
+        # (consumed by the next block's fork-conv dgrad epilogue)
         if (self.fuse_bwd_stats and _FUSE_BNBWD and self.training
-                and x.is_cuda and relu and residual is None
+                and x.is_cuda and relu
                 and x.dtype != torch.float32
                 and torch.is_grad_enabled()):
             holder = DF.BnBwdFusion()

@@ -78,3 +78,48 @@ def test_fused_matches_fp32_reference(monkeypatch):
         scale = refv.abs().max().clamp_min(1e-6)
         diff = (mine - refv).abs().max() / scale
         assert diff < 5e-2, (name, float(diff))
+
+
+def _bottleneck_grads(enabled: bool):
+    """Full Bottleneck chain: exercises BOTH fusion forms — bn1/bn2
+    (recomputed mask via the chain conv) and bn3 (saved bit-mask via the
+    next block's fork-conv accum epilogue)."""
+    import distributeddeeplearning_amd.ops.modules as M
+    import distributeddeeplearning_amd.ops.functional as DF
+    from distributeddeeplearning_amd.models.resnet import Bottleneck
+    old = M._FUSE_BNBWD
+    M._FUSE_BNBWD = enabled
+    DF.set_direct_wgrad(None)
+    try:
+        torch.manual_seed(11)
+        b1 = Bottleneck(64, 16).cuda().train()
+        b2 = Bottleneck(64, 16).cuda().train()
+        for blk in (b1, b2):
+            for p in blk.parameters():
+                p.data = (cl(p.data.to(torch.bfloat16)) if p.dim() == 4
+                          else p.data.to(torch.bfloat16))
+        x = cl(torch.randn(4, 64, 14, 14, device="cuda",
+                           dtype=torch.bfloat16) * 0.5)
+        x.requires_grad_(True)
+        # two chained blocks: b1's bn3 output is consumed by b2's fork
+        y = b2(b1(x))
+        (y.float() ** 2).mean().backward()
+        torch.cuda.synchronize()
+        g = {n: p.grad.detach().float().cpu().clone()
+             for blk, tag in ((b1, "b1"), (b2, "b2"))
+             for n, p in blk.named_parameters()
+             for n in (f"{tag}.{n}",)}
+        g["dx"] = x.grad.detach().float().cpu().clone()
+        return g
+    finally:
+        M._FUSE_BNBWD = old
+
+
+def test_bottleneck_chain_fused_matches_unfused():
+    a = _bottleneck_grads(True)
+    b = _bottleneck_grads(False)
+    assert a.keys() == b.keys()
+    for k in a:
+        scale = b[k].abs().max().clamp_min(1e-6)
+        diff = (a[k] - b[k]).abs().max() / scale
+        assert diff < 3e-2, (k, float(diff))
