@@ -263,7 +263,7 @@ class _ForkConvFn(torch.autograd.Function):
             if dst is not None:
                 _grad_direct.commit(weight)
                 dw = None
-        return dx, dw, None, None, None
+        return dx, dw, None, None, None, None
 
 
 _FUSE_RESJOIN = os.environ.get("DDL_FUSE_RESJOIN", "1") == "1"
