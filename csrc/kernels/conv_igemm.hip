@@ -776,8 +776,8 @@ void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
     const int64_t grid =
         n64 ? cdiv(M, 256) * cdiv(Nc, 64) : cdiv(M, BM) * cdiv(Nc, BN);
     DgradPhase gp{};
-#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD, false, 128, 128, true>                                      <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0)
-#define RUNB64(T)                                                                conv_igemm_k<T, DGRAD, false, 256, 64, true>                                       <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0)
+#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD, false, 128, 128, true>                                      <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, (const T*)accum, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0, bn_mask)
+#define RUNB64(T)                                                                conv_igemm_k<T, DGRAD, false, 256, 64, true>                                       <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, (const T*)accum, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, 0, bn_mask)
     if (dtype == 1) { if (n64) RUNB64(bf16); else RUNB(bf16); }
     else { if (n64) RUNB64(f16); else RUNB(f16); }
 #undef RUNB64
@@ -799,7 +799,7 @@ void ddl_conv_dgrad_bnstats(const void* dy, const void* w_t, void* dx,
       gp.fdHpWp = make_fastdiv((uint32_t)gp.Hp * gp.Wp);
       const int64_t M = (int64_t)N * gp.Hp * gp.Wp;
       const int64_t grid = cdiv(M, BM) * cdiv(Nc, BN);
-#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD_S2, false, 128, 128, true>                                   <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, nullptr, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, chunk_base)
+#define RUNB(T)                                                                  conv_igemm_k<T, DGRAD_S2, false, 128, 128, true>                                   <<<(unsigned)grid, 256, 0, s>>>(                                                   (const T*)dy, (const T*)w_t, nullptr, (T*)dx, d, M, Nc, 0, gp,                 nullptr, (const T*)accum, (const T*)bnx, bn_mean, bn_rstd, (const T*)bn_w,             (const T*)bn_b, bstats_ws, chunk_base, bn_mask)
       if (dtype == 1) RUNB(bf16); else RUNB(f16);
 #undef RUNB
       chunk_base += (int)cdiv(M, BM);
