@@ -12,7 +12,12 @@ from __future__ import annotations
 
 from typing import List, Type, Union
 
+import os
+
 import torch.nn as nn
+
+# residual-join BN backward-stats fusion (measured -2% e2e; opt-in)
+_FUSE_BN3 = os.environ.get("DDL_FUSE_BN3", "0")
 
 from ..ops import functional as DF
 from ..ops.modules import (
@@ -44,7 +49,7 @@ class BasicBlock(nn.Module):
         # stats come from conv2's dgrad epilogue (BnBwdFusion)
         self.bn1 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=True)  # join BN (BasicBlock)
+        self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=_FUSE_BN3 == "1")  # join BN
         self.downsample = downsample
         self.stride = stride
 
@@ -72,11 +77,13 @@ class Bottleneck(nn.Module):
         self.conv2 = conv3x3(planes, planes, stride)
         self.bn2 = BatchNorm2d(planes, fuse_bwd_stats=True)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        # bn3's join output has ONE autograd consumer too (the next
-        # block's fork-conv, which carries the identity grad through its
-        # accum epilogue) — its backward stats fuse there, masked by the
-        # forward's bit-packed ReLU mask
-        self.bn3 = BatchNorm2d(planes * self.expansion, fuse_bwd_stats=True)
+        # bn3's backward-stats CAN fuse into the next block's fork-conv
+        # accum epilogue (mask mode; capability tested in
+        # test_bnbwd_fusion_gpu), but measured -2% end-to-end: the extra
+        # epilogue reads sit on the serial dgrad path while the stats
+        # kernels they replace overlapped well.  Default off.
+        self.bn3 = BatchNorm2d(planes * self.expansion,
+                               fuse_bwd_stats=_FUSE_BN3 == "1")
         self.downsample = downsample
         self.stride = stride
 
