@@ -319,6 +319,8 @@ class Trainer:
             if torch_dist.is_initialized():
                 torch_dist.all_reduce(t, op=torch_dist.ReduceOp.MAX)
             elapsed = float(t.item())
+        if self.timeline:
+            self.timeline.save()  # flush all spans for overlap analysis
         ms_per_step = elapsed / steps * 1000.0
         images_per_sec = self.global_batch * steps / elapsed
         return {"images_per_sec": images_per_sec, "ms_per_step": ms_per_step,
