@@ -35,7 +35,12 @@ class _DirectGradRegistry:
     window (later passes must take the autograd-accumulate path so the
     running sum is preserved); `commit` replays exactly what the
     post-accumulate-grad hook would have done — stall timestamp, delay
-    decrement, bucket-ready bookkeeping, collective launch."""
+    decrement, bucket-ready bookkeeping, collective launch.
+
+    Caveat: a conv weight REUSED at several graph sites would commit at
+    its first use, before later contributions land — set
+    DDL_DIRECT_WGRAD=0 for weight-sharing models (none of the in-repo
+    model families share conv weights)."""
 
     def __init__(self, opt: "DistributedOptimizer"):
         self.opt = opt
