@@ -5,10 +5,8 @@ over-backward detection, fp16-overflow recovery mid-accumulation, exact
 fp32-master restore under low-precision params, and CPU-metric reduction
 routing.
 """
-import copy
 import os
 
-import pytest
 import torch
 
 from tests.dist_helpers import run_distributed

@@ -29,7 +29,7 @@ def main():
     p.add_argument("--out", default="gpurun_out/lars_curve.json")
     args = p.parse_args()
 
-    import torch
+    import torch  # noqa: F401 — CUDA init side effect
     from distributeddeeplearning_amd.config import PRESETS
     from distributeddeeplearning_amd.engine.trainer import Trainer
 
