@@ -257,6 +257,7 @@ class Trainer:
         return (os.environ.get("DDL_HIPGRAPH", "0") == "1"
                 and self.device.type == "cuda" and self.world == 1
                 and not self.scaler.enabled
+                and self.cfg.data == "synthetic"  # loader must CYCLE
                 and self.cfg.model.startswith("resnet"))
 
     def _capture_graphs(self, it) -> list:
@@ -269,8 +270,8 @@ class Trainer:
         graphs = []
         seen = set()
         torch.cuda.synchronize(self.device)
-        while True:
-            x, y = next(it)
+        while len(graphs) < 16:  # hard cap: a non-cycling loader must not
+            x, y = next(it)      # capture unboundedly
             if x.data_ptr() in seen:
                 break
             seen.add(x.data_ptr())
