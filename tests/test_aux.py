@@ -146,3 +146,39 @@ def test_example_script_cpu_smoke(tmp_path):
         env={**os.environ, "DDL_COMM": "single"})
     assert r.returncode == 0, r.stderr[-800:]
     assert (tmp_path / "resnet50_epoch1.pt").exists()
+
+
+def test_scale8_overlap_parser(tmp_path):
+    """tools/scale8.py's timeline-overlap analysis: RCCL spans inside the
+    hipEvent-timed backward window are measured correctly."""
+    import json
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "scale8", os.path.join(os.path.dirname(__file__), "..", "tools",
+                               "scale8.py"))
+    scale8 = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(scale8)
+
+    host = {"traceEvents": [
+        {"name": "dev/backward+allreduce", "ph": "X", "ts": 1000.0,
+         "dur": 4000.0, "pid": 0, "tid": 1},
+        {"name": "forward", "ph": "X", "ts": 0.0, "dur": 1000.0},
+    ]}
+    comm = {"traceEvents": [
+        # 2000us span, 1500 inside the backward window
+        {"name": "span/allreduce#1/25 MB", "ph": "X", "ts": 3500.0,
+         "dur": 2000.0, "pid": 0, "tid": 1},
+        # fully inside
+        {"name": "span/allreduce#2/25 MB", "ph": "X", "ts": 2000.0,
+         "dur": 1000.0, "pid": 0, "tid": 1},
+        # not an allreduce span: ignored
+        {"name": "enqueue/broadcast#0/1 MB", "ph": "X", "ts": 0.0,
+         "dur": 50.0},
+    ]}
+    pre = str(tmp_path / "tl")
+    json.dump(host, open(pre + ".rank0.json", "w"))
+    json.dump(comm, open(pre + ".rank0.comm.json", "w"))
+    res = scale8.overlap_from_timelines(pre, 1)
+    assert res["rank0"]["rccl_span_ms"] == 3.0
+    assert res["rank0"]["inside_backward_ms"] == 2.5
+    assert abs(res["rank0"]["overlap_frac"] - 2.5 / 3.0) < 1e-3  # rounded to 3 places
