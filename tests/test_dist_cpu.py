@@ -219,3 +219,32 @@ def test_world4_with_grad_accumulation():
     res = run_distributed(_w_world4, world=4, timeout=300)
     for r in res[1:]:
         assert torch.equal(res[0]["params"], r["params"])
+
+
+def _w_dtype_matrix(rank, world):
+    """Horovod-suite theme (test_torch.py): allreduce/broadcast across
+    dtypes and shapes, including 0-dim and large tensors."""
+    import torch
+    from distributeddeeplearning_amd.parallel import api
+    api.init("torchdist")
+    out = {}
+    for dt in (torch.float32, torch.float64, torch.int64):
+        for shape in ((), (1,), (7, 3), (100_000,)):
+            t = torch.ones(shape, dtype=dt) * (rank + 1)
+            r = api.allreduce(t, average=False)
+            out[f"ar/{dt}/{shape}"] = bool((r == world * (world + 1) // 2).all())
+            b = torch.full(shape, float(rank)).to(dt)
+            api.broadcast_(b, root_rank=0)
+            out[f"bc/{dt}/{shape}"] = bool((b == 0).all())
+    # average on float dtypes
+    t = torch.full((5,), float(rank + 1))
+    r = api.allreduce(t, average=True)
+    out["avg"] = bool((r == (world + 1) / 2).all())
+    return out
+
+
+def test_collective_dtype_shape_matrix():
+    res = run_distributed(_w_dtype_matrix, world=2)
+    for r in res:
+        bad = [k for k, v in r.items() if not v]
+        assert not bad, bad
