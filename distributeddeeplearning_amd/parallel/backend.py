@@ -97,7 +97,10 @@ class TorchDistBackend(CommBackend):
         return torch_dist.get_world_size(self.pg)
 
     def allreduce_async_(self, tensor: torch.Tensor, average: bool = True) -> Handle:
-        work = torch_dist.all_reduce(tensor, op=torch_dist.ReduceOp.SUM, group=self.pg, async_op=True)
+        # gloo aborts on 0-dim tensors; hvd supports them — reduce through
+        # a storage-sharing 1-element view
+        wire = tensor.unsqueeze(0) if tensor.dim() == 0 else tensor
+        work = torch_dist.all_reduce(wire, op=torch_dist.ReduceOp.SUM, group=self.pg, async_op=True)
         post = None
         if average and self.size() > 1:
             n = self.size()
@@ -108,7 +111,8 @@ class TorchDistBackend(CommBackend):
         return Handle(work, tensor, post)
 
     def broadcast_async_(self, tensor: torch.Tensor, root_rank: int) -> Handle:
-        work = torch_dist.broadcast(tensor, src=root_rank, group=self.pg, async_op=True)
+        wire = tensor.unsqueeze(0) if tensor.dim() == 0 else tensor
+        work = torch_dist.broadcast(wire, src=root_rank, group=self.pg, async_op=True)
         return Handle(work, tensor)
 
     def allgather_async(self, tensor: torch.Tensor) -> Handle:
