@@ -70,3 +70,25 @@ def test_checkpoint_periodic_and_resume(tmp_path):
     cfg2 = _cfg(resume=str(ck))
     tr2 = Trainer(cfg2, steps_per_epoch=2)
     assert tr2.epoch == 1 and tr2.global_step == 2
+
+
+def test_eval_cli_roundtrip(tmp_path):
+    """tools/eval.py end-to-end: train 2 steps, checkpoint, evaluate."""
+    import json
+    import os
+    import subprocess
+    import sys
+    cfg = _cfg(checkpoint_dir=str(tmp_path), epochs=1, steps=2)
+    tr = Trainer(cfg, steps_per_epoch=2)
+    tr.fit()
+    ck = tmp_path / "resnet18_epoch1.pt"
+    root = os.path.join(os.path.dirname(__file__), "..")
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "tools", "eval.py"),
+         "--checkpoint", str(ck), "--model", "resnet18",
+         "--image-size", "32", "--num-classes", "10",
+         "--batch-size", "8", "--max-batches", "2"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert out.returncode == 0, out.stderr[-1500:]
+    rec = json.loads(out.stdout.strip().splitlines()[-1])
+    assert 0.0 <= rec["top1"] <= 1.0 and rec["val_loss"] == rec["val_loss"]
