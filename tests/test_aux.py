@@ -182,3 +182,17 @@ def test_scale8_overlap_parser(tmp_path):
     assert res["rank0"]["rccl_span_ms"] == 3.0
     assert res["rank0"]["inside_backward_ms"] == 2.5
     assert abs(res["rank0"]["overlap_frac"] - 2.5 / 3.0) < 1e-3  # rounded to 3 places
+
+
+def test_example_script_argparse():
+    """examples/imagenet_resnet50.py (the call-for-call reference port)
+    stays importable and its CLI parses."""
+    import subprocess
+    import sys
+    root = os.path.join(os.path.dirname(__file__), "..")
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "examples",
+                                      "imagenet_resnet50.py"), "--help"],
+        capture_output=True, text=True, timeout=120, cwd=root)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "--batch-size" in out.stdout
