@@ -27,21 +27,46 @@ import torch
 import torch.distributed as torch_dist
 
 
+_STALL_WARN_S = float(os.environ.get("DDL_STALL_WARN_SECONDS", "60"))
+
+
 class Handle:
     """An async-op handle: backend-specific payload plus the target tensor."""
 
-    __slots__ = ("work", "tensor", "post", "_done")
+    __slots__ = ("work", "tensor", "post", "_done", "stall_guard")
 
-    def __init__(self, work: Any, tensor: torch.Tensor, post=None):
+    def __init__(self, work: Any, tensor: torch.Tensor, post=None,
+                 stall_guard: bool = False):
         self.work = work
         self.tensor = tensor
         self.post = post  # optional callable run once at synchronize()
         self._done = False
+        # poll-with-warning instead of a silent indefinite block (gloo
+        # control/data plane only — NCCL work keeps its efficient
+        # stream-semantics wait)
+        self.stall_guard = stall_guard
+
+    def _guarded_wait(self) -> None:
+        import time as _time
+        import warnings as _warnings
+        t0 = _time.monotonic()
+        warned = False
+        while not self.work.is_completed():
+            if not warned and _time.monotonic() - t0 > _STALL_WARN_S:
+                _warnings.warn(
+                    f"collective outstanding for {_STALL_WARN_S:.0f}s on the "
+                    "torch.distributed path — possible rank desync/deadlock "
+                    "(stall inspector)")
+                warned = True
+            _time.sleep(0.002)
 
     def wait(self) -> torch.Tensor:
         if not self._done:
             if self.work is not None:
-                self.work.wait()
+                if self.stall_guard and hasattr(self.work, "is_completed"):
+                    self._guarded_wait()
+                else:
+                    self.work.wait()
             if self.post is not None:
                 self.post()
             self._done = True
@@ -108,12 +133,14 @@ class TorchDistBackend(CommBackend):
             def post(t=tensor, n=n):
                 t.div_(n)
 
-        return Handle(work, tensor, post)
+        # per-tensor: CUDA collectives keep NCCL's efficient stream-
+        # semantics wait; host-side (gloo) waits get the stall warning
+        return Handle(work, tensor, post, stall_guard=not tensor.is_cuda)
 
     def broadcast_async_(self, tensor: torch.Tensor, root_rank: int) -> Handle:
         wire = tensor.unsqueeze(0) if tensor.dim() == 0 else tensor
         work = torch_dist.broadcast(wire, src=root_rank, group=self.pg, async_op=True)
-        return Handle(work, tensor)
+        return Handle(work, tensor, stall_guard=not tensor.is_cuda)
 
     def allgather_async(self, tensor: torch.Tensor) -> Handle:
         ws = self.size()
