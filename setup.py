@@ -16,7 +16,7 @@ import sys
 from concurrent.futures import ThreadPoolExecutor
 from pathlib import Path
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -92,7 +92,8 @@ ext_modules = [
 setup(
     name="distributeddeeplearning_amd",
     version="0.1.0",
-    packages=["distributeddeeplearning_amd"],
+    packages=find_packages(include=["distributeddeeplearning_amd",
+                                    "distributeddeeplearning_amd.*"]),
     ext_modules=ext_modules,
     cmdclass={"build_ext": BuildExtension},
 )
