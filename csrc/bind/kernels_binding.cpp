@@ -632,9 +632,12 @@ void fused_sgd(at::Tensor table, c10::optional<at::Tensor> norms,
 
 void lars_norms(at::Tensor table, at::Tensor norms, int64_t dtype, double wd,
                 double inv_scale) {
+  // deterministic two-stage: per-chunk partials then a fixed-order fold
+  auto partials = at::empty({2 * table.size(0)},
+                            norms.options().dtype(at::kFloat));
   ddl_lars_norms(table.data_ptr<int64_t>(), table.size(0),
-                 norms.data_ptr<float>(), norms.size(1), dtype, wd, inv_scale,
-                 cur_stream());
+                 norms.data_ptr<float>(), partials.data_ptr<float>(),
+                 norms.size(1), dtype, wd, inv_scale, cur_stream());
 }
 
 }  // namespace

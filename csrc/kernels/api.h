@@ -30,9 +30,11 @@ void ddl_fused_sgd(const int64_t* table, int nrows, const float* norms,
                    int ntensors, int dtype, float lr, float momentum,
                    float wd, int nesterov, float lars_eta, float lars_eps,
                    float inv_scale, hipStream_t s);
+// deterministic two-stage: per-chunk partials[2*nrows] then a per-tensor
+// fold in fixed chunk order (no atomics)
 void ddl_lars_norms(const int64_t* table, int nrows, float* norms,
-                    int ntensors, int dtype, float wd, float inv_scale,
-                    hipStream_t s);
+                    float* partials, int ntensors, int dtype, float wd,
+                    float inv_scale, hipStream_t s);
 
 // ---- fused BatchNorm + residual + ReLU (K4/K5/K14) ------------------------
 // sums: f32 [2*C]; partials: f32 [chunks*2*C] scratch where chunks =

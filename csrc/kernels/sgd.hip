@@ -54,32 +54,52 @@ __global__ void fused_sgd_k(const int64_t* table, const float* norms,
   }
 }
 
-// Per-chunk partial squared norms, atomically accumulated per tensor:
-// norms[ti] += sum(master^2), norms[NT + ti] += sum(g'^2),
-// g' = grad*inv_scale + wd*master  (the effective gradient LARS scales).
+// Per-chunk partial squared norms -> partials[2][nrows] (deterministic,
+// no atomics): sw = sum(master^2), sg = sum(g'^2) with
+// g' = grad*inv_scale + wd*master (the effective gradient LARS scales).
 template <typename T>
-__global__ void lars_norms_k(const int64_t* table, float* norms, int ntensors,
+__global__ void lars_norms_k(const int64_t* table, float* partials, int nrows,
                              float wd, float inv_scale) {
   __shared__ float scratch[8];
   const Row row = *reinterpret_cast<const Row*>(table + (int64_t)blockIdx.x * 7);
-  if (!row.lars) return;
-  const T* grad = reinterpret_cast<const T*>(row.g);
-  const float* master = reinterpret_cast<const float*>(row.m);
-  const int n = (int)row.n;
   float sw = 0.f, sg = 0.f;
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    float w = master[i];
-    float g = to_f32(grad[i]) * inv_scale + wd * w;
-    sw += w * w;
-    sg += g * g;
+  if (row.lars) {
+    const T* grad = reinterpret_cast<const T*>(row.g);
+    const float* master = reinterpret_cast<const float*>(row.m);
+    const int n = (int)row.n;
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      float w = master[i];
+      float g = to_f32(grad[i]) * inv_scale + wd * w;
+      sw += w * w;
+      sg += g * g;
+    }
   }
   sw = block_reduce_sum(sw, scratch);
   __syncthreads();
   sg = block_reduce_sum(sg, scratch);
   if (threadIdx.x == 0) {
-    atomicAdd(&norms[row.ti], sw);
-    atomicAdd(&norms[ntensors + row.ti], sg);
+    partials[blockIdx.x] = sw;
+    partials[nrows + blockIdx.x] = sg;
   }
+}
+
+// Fold the per-chunk partials into per-tensor norms in fixed chunk order:
+// one block per tensor scans the (tensor-sorted) rows.
+__global__ void lars_fold_norms_k(const int64_t* table,
+                                  const float* partials, float* norms,
+                                  int nrows, int ntensors) {
+  const int ti = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  float sw = 0.f, sg = 0.f;
+  for (int r = 0; r < nrows; ++r) {
+    const Row row = *reinterpret_cast<const Row*>(table + (int64_t)r * 7);
+    if ((int)row.ti == ti && row.lars) {
+      sw += partials[r];
+      sg += partials[nrows + r];
+    }
+  }
+  norms[ti] = sw;
+  norms[ntensors + ti] = sg;
 }
 
 }  // namespace
@@ -99,12 +119,14 @@ void ddl_fused_sgd(const int64_t* table, int nrows, const float* norms,
 }
 
 void ddl_lars_norms(const int64_t* table, int nrows, float* norms,
-                    int ntensors, int dtype, float wd, float inv_scale,
-                    hipStream_t s) {
+                    float* partials, int ntensors, int dtype, float wd,
+                    float inv_scale, hipStream_t s) {
 #define RUN(T)                                                                 \
-  lars_norms_k<T><<<nrows, 256, 0, s>>>(table, norms, ntensors, wd, inv_scale)
+  lars_norms_k<T><<<nrows, 256, 0, s>>>(table, partials, nrows, wd, inv_scale)
   switch (dtype) { case 0: RUN(float); break; case 1: RUN(bf16); break; case 2: RUN(f16); break; }
 #undef RUN
+  lars_fold_norms_k<<<ntensors, 32, 0, s>>>(table, partials, norms, nrows,
+                                            ntensors);
 }
 
 }  // extern "C"

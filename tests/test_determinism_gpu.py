@@ -14,12 +14,12 @@ import torch
 pytestmark = pytest.mark.gpu
 
 
-def _run(steps=8, dtype="bf16", model="resnet50"):
+def _run(steps=8, dtype="bf16", model="resnet50", lars=False):
     from distributeddeeplearning_amd.config import TrainConfig
     from distributeddeeplearning_amd.engine.trainer import Trainer
     cfg = TrainConfig(model=model, image_size=64, num_classes=100,
                       batch_size=32, dtype=dtype, warmup_epochs=0,
-                      base_lr=0.05, comm="single", seed=123)
+                      base_lr=0.05, comm="single", seed=123, lars=lars)
     tr = Trainer(cfg, steps_per_epoch=10 ** 9)
     it = tr.data_iter()
     for _ in range(steps):
@@ -35,10 +35,11 @@ def _run(steps=8, dtype="bf16", model="resnet50"):
     return params, stats, float(loss.detach().float())
 
 
-@pytest.mark.parametrize("dtype", ["bf16", "fp32"])
-def test_training_is_bit_deterministic(dtype):
-    p1, s1, l1 = _run(dtype=dtype)
-    p2, s2, l2 = _run(dtype=dtype)
+@pytest.mark.parametrize("dtype,lars", [("bf16", False), ("fp32", False),
+                                        ("bf16", True)])
+def test_training_is_bit_deterministic(dtype, lars):
+    p1, s1, l1 = _run(dtype=dtype, lars=lars)
+    p2, s2, l2 = _run(dtype=dtype, lars=lars)
     assert l1 == l2
     assert torch.equal(p1, p2), "parameters diverged between identical runs"
     assert torch.equal(s1, s2), "BN running stats diverged"
