@@ -48,6 +48,7 @@ class TrainConfig:
     comm: str = "auto"               # auto | native | torchdist | single
     bucket_mb: float = 25.0
     backward_passes_per_step: int = 1
+    autotune_buckets: bool = False   # online bucket-size tuner (DDL_AUTOTUNE)
 
     # engine
     device: str = "auto"             # auto | cpu | cuda
@@ -127,6 +128,8 @@ class TrainConfig:
             cfg.bucket_mb = float(os.environ["DDL_BUCKET_MB"])
         if "DDL_COMM" in os.environ:
             cfg.comm = os.environ["DDL_COMM"]
+        if os.environ.get("DDL_AUTOTUNE") == "1":
+            cfg.autotune_buckets = True
         return cfg
 
 

@@ -103,6 +103,14 @@ class Trainer:
 
         self._loader: Optional[Iterator] = None
         self._micro = 0  # backward passes since the last optimizer step
+        # online bucket autotune (Horovod parameter_manager role): walks a
+        # candidate ladder during early training, every rank picks the
+        # allreduced winner, buckets rebuild once
+        self.bucket_tuner = None
+        if (cfg.autotune_buckets
+                or os.environ.get("DDL_AUTOTUNE") == "1"):
+            from ..parallel.autotune import OnlineBucketTuner
+            self.bucket_tuner = OnlineBucketTuner(self.opt)
 
     # ------------------------------------------------------------------
 
@@ -169,6 +177,8 @@ class Trainer:
         else:
             self.opt.step()
         self.opt.zero_grad()
+        if self.bucket_tuner is not None and not self.bucket_tuner.done:
+            self.bucket_tuner.step()
         if tl:
             tl.end("step")
             tl.end_device("step")

@@ -279,6 +279,34 @@ class DistributedOptimizer:
             "supported: bucket membership is static"
         )
 
+    def rebucket(self, bucket_bytes: int) -> None:
+        """Rebuild the buckets at a new size (online autotune): new flat
+        buffers + grad views, hooks re-registered, delay counters re-armed,
+        direct-wgrad registry refreshed.  Must run at an optimizer-step
+        boundary, at the SAME step on every rank (the collective order
+        changes with the bucket layout)."""
+        for h in self._hooks:
+            h.remove()
+        named = list(zip(self.buckets.names[::-1],
+                         [p for b in self.buckets.buckets for p in b.params]))
+        # names list is reverse-registration order matching the flattened
+        # bucket params; restore registration order for the new manager
+        named = named[::-1]
+        self.buckets = BucketManager(named, bucket_bytes=bucket_bytes)
+        self._hooks = []
+        self._delay = {}
+        for b in self.buckets.buckets:
+            for p in b.params:
+                self._delay[id(p)] = self.backward_passes_per_step
+                h = p.register_post_accumulate_grad_hook(self._make_hook())
+                self._hooks.append(h)
+        self._synchronized = False
+        if _DIRECT_WGRAD:
+            from ..ops import functional as DF
+            DF.set_direct_wgrad(_DirectGradRegistry(self))
+        if hasattr(self.optimizer, "invalidate_table"):
+            self.optimizer.invalidate_table()  # grad pointers changed
+
     def set_timeline(self, timeline):
         self._timeline = timeline
 

@@ -248,3 +248,34 @@ def test_collective_dtype_shape_matrix():
     for r in res:
         bad = [k for k, v in r.items() if not v]
         assert not bad, bad
+
+
+def _w_autotune(rank, world):
+    """Online bucket autotune: every rank walks the same candidate ladder,
+    picks the allreduced winner, and the ranks stay bit-identical."""
+    import dataclasses
+    from distributeddeeplearning_amd.config import PRESETS
+    from distributeddeeplearning_amd.engine.trainer import Trainer
+    cfg = dataclasses.replace(PRESETS["cpu-plumbing"], batch_size=8,
+                              autotune_buckets=True)
+    tr = Trainer(cfg, steps_per_epoch=100)
+    tr.bucket_tuner.sizes = [0.05, 0.2]          # tiny ladder for CI
+    tr.bucket_tuner.steps_per = 3
+    tr.bucket_tuner.warmup = 1
+    it = tr.data_iter()
+    for _ in range(10):
+        x, y = next(it)
+        tr.train_step(x, y)
+    params = torch.cat([p.detach().reshape(-1)
+                        for p in tr.model.parameters()])
+    return {"params": params, "done": tr.bucket_tuner.done,
+            "chosen": tr.bucket_tuner.chosen_mb,
+            "n_buckets": len(tr.opt.buckets.buckets)}
+
+
+def test_online_bucket_autotune_world2():
+    res = run_distributed(_w_autotune, world=2, timeout=300)
+    assert all(r["done"] for r in res)
+    assert res[0]["chosen"] == res[1]["chosen"], "ranks chose different sizes"
+    assert res[0]["n_buckets"] == res[1]["n_buckets"]
+    assert torch.equal(res[0]["params"], res[1]["params"]), "ranks diverged"
